@@ -1,0 +1,180 @@
+// Torch bindings for the fluxdistributed_amd gfx950 kernels (_C extension).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "fda_kernels.h"
+
+namespace {
+
+using fda::DT;
+
+DT dt_of(const at::Tensor& t) {
+    if (t.scalar_type() == at::kBFloat16) return DT::BF16;
+    TORCH_CHECK(t.scalar_type() == at::kFloat, "expected f32 or bf16, got ",
+                t.scalar_type());
+    return DT::F32;
+}
+
+hipStream_t cur_stream() {
+    return at::hip::getCurrentHIPStream().stream();
+}
+
+// channels_last 4D tensor -> (rows, C) with C contiguous
+std::pair<int64_t, int64_t> nhwc_rows(const at::Tensor& x) {
+    TORCH_CHECK(x.dim() == 4, "expected 4D NCHW-logical tensor");
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "expected channels_last layout");
+    return {x.size(0) * x.size(2) * x.size(3), x.size(1)};
+}
+
+std::tuple<at::Tensor, at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target) {
+    TORCH_CHECK(logits.dim() == 2 && logits.is_contiguous());
+    TORCH_CHECK(target.scalar_type() == at::kLong && target.is_contiguous());
+    const int N = (int)logits.size(0), C = (int)logits.size(1);
+    auto loss = at::zeros({}, logits.options().dtype(at::kFloat));
+    auto dlogits = at::empty_like(logits);
+    fda::ce_fwd_launch(logits.data_ptr(), target.data_ptr<int64_t>(),
+                       loss.data_ptr<float>(), dlogits.data_ptr(), N, C,
+                       dt_of(logits), cur_stream());
+    return {loss.to(logits.scalar_type()), dlogits};
+}
+
+at::Tensor add_relu_fwd(at::Tensor x, at::Tensor r) {
+    TORCH_CHECK(x.sizes() == r.sizes() && x.scalar_type() == r.scalar_type());
+    at::Tensor xc = x, rc = r;
+    if (x.dim() == 4) {
+        xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+        rc = r.contiguous(at::MemoryFormat::ChannelsLast);
+    } else {
+        xc = x.contiguous();
+        rc = r.contiguous();
+    }
+    auto out = at::empty_like(xc);
+    fda::add_relu_fwd_launch(xc.data_ptr(), rc.data_ptr(), out.data_ptr(),
+                             xc.numel(), dt_of(xc), cur_stream());
+    return out;
+}
+
+at::Tensor add_relu_bwd(at::Tensor gout, at::Tensor out) {
+    TORCH_CHECK(gout.sizes() == out.sizes());
+    auto gc = gout, oc = out;
+    if (gout.dim() == 4) {
+        gc = gout.contiguous(at::MemoryFormat::ChannelsLast);
+        oc = out.contiguous(at::MemoryFormat::ChannelsLast);
+    } else {
+        gc = gout.contiguous();
+        oc = out.contiguous();
+    }
+    auto gx = at::empty_like(gc);
+    fda::add_relu_bwd_launch(gc.data_ptr(), oc.data_ptr(), gx.data_ptr(),
+                             gc.numel(), dt_of(gc), cur_stream());
+    return gx;
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
+    at::Tensor x, at::Tensor weight, at::Tensor bias, at::Tensor running_mean,
+    at::Tensor running_var, bool training, double momentum, double eps,
+    bool relu, at::Tensor residual) {
+    auto [rows, C] = nhwc_rows(x);
+    const int V = dt_of(x) == DT::BF16 ? 8 : 4;
+    TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
+    TORCH_CHECK((C / V) >= 256 ? (C / V) % 256 == 0 : 256 % (C / V) == 0,
+                "unsupported channel count ", C);
+    TORCH_CHECK(weight.scalar_type() == at::kFloat, "BN params must be fp32");
+    const bool has_res = residual.defined() && residual.numel() > 0;
+    at::Tensor resc;
+    if (has_res) resc = residual.contiguous(at::MemoryFormat::ChannelsLast);
+
+    auto fopts = x.options().dtype(at::kFloat);
+    auto ws = at::empty({4 * C}, fopts);
+    auto save_mean = at::empty({C}, fopts);
+    auto save_invstd = at::empty({C}, fopts);
+    auto out = at::empty_like(x);
+    auto stream = cur_stream();
+
+    if (training) {
+        ws.narrow(0, 0, 2 * C).zero_();
+        fda::bn_stats_launch(x.data_ptr(), ws.data_ptr<float>(), rows, (int)C,
+                             dt_of(x), stream);
+    }
+    fda::bn_finalize_launch(ws.data_ptr<float>(), weight.data_ptr<float>(),
+                            bias.data_ptr<float>(),
+                            running_mean.data_ptr<float>(),
+                            running_var.data_ptr<float>(),
+                            save_mean.data_ptr<float>(),
+                            save_invstd.data_ptr<float>(), rows, (int)C,
+                            training, (float)momentum, (float)eps, stream);
+    fda::bn_apply_launch(x.data_ptr(), has_res ? resc.data_ptr() : nullptr,
+                         out.data_ptr(), ws.data_ptr<float>(), rows, (int)C,
+                         relu, dt_of(x), stream);
+    return {out, save_mean, save_invstd};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
+    at::Tensor gout, at::Tensor x, at::Tensor weight, at::Tensor save_mean,
+    at::Tensor save_invstd, at::Tensor out, bool relu, bool training) {
+    auto [rows, C] = nhwc_rows(x);
+    auto gc = gout.contiguous(at::MemoryFormat::ChannelsLast);
+    auto fopts = x.options().dtype(at::kFloat);
+    auto ws = at::zeros({4 * C}, fopts);
+    auto gw = at::empty({C}, fopts);
+    auto gb = at::empty({C}, fopts);
+    auto gx = at::empty_like(x);
+    auto stream = cur_stream();
+
+    fda::bn_bwd_stats_launch(gc.data_ptr(), x.data_ptr(), out.data_ptr(),
+                             save_mean.data_ptr<float>(),
+                             save_invstd.data_ptr<float>(), ws.data_ptr<float>(),
+                             rows, (int)C, relu, dt_of(x), stream);
+    fda::bn_bwd_finalize_launch(ws.data_ptr<float>(), weight.data_ptr<float>(),
+                                save_invstd.data_ptr<float>(),
+                                gw.data_ptr<float>(), gb.data_ptr<float>(),
+                                rows, (int)C, training, stream);
+    fda::bn_bwd_apply_launch(gc.data_ptr(), x.data_ptr(), out.data_ptr(),
+                             save_mean.data_ptr<float>(),
+                             save_invstd.data_ptr<float>(),
+                             weight.data_ptr<float>(), ws.data_ptr<float>(),
+                             gx.data_ptr(), rows, (int)C, relu, training,
+                             dt_of(x), stream);
+    return {gx, gw, gb};
+}
+
+void sgd_step(at::Tensor P, at::Tensor G, at::Tensor M, at::Tensor V,
+              double lr, double mom, double wd, bool nesterov) {
+    TORCH_CHECK(P.is_contiguous() && G.is_contiguous() && V.is_contiguous());
+    const int vw = dt_of(P) == DT::BF16 ? 8 : 4;
+    TORCH_CHECK(P.numel() % vw == 0, "flat buffer must be padded to ", vw);
+    const bool has_master = M.data_ptr() != P.data_ptr();
+    if (dt_of(P) == DT::BF16) TORCH_CHECK(has_master, "bf16 params need fp32 master");
+    fda::sgd_step_launch(P.data_ptr(), G.data_ptr(), M.data_ptr<float>(),
+                         V.data_ptr<float>(), P.numel(), (float)lr, (float)mom,
+                         (float)wd, nesterov, has_master, dt_of(P),
+                         cur_stream());
+}
+
+void adam_step(at::Tensor P, at::Tensor G, at::Tensor M, at::Tensor V,
+               at::Tensor S, double lr, double b1, double b2, double eps,
+               double wd, double bc1, double bc2) {
+    const int vw = dt_of(P) == DT::BF16 ? 8 : 4;
+    TORCH_CHECK(P.numel() % vw == 0, "flat buffer must be padded to ", vw);
+    const bool has_master = M.data_ptr() != P.data_ptr();
+    if (dt_of(P) == DT::BF16) TORCH_CHECK(has_master, "bf16 params need fp32 master");
+    fda::adam_step_launch(P.data_ptr(), G.data_ptr(), M.data_ptr<float>(),
+                          V.data_ptr<float>(), S.data_ptr<float>(), P.numel(),
+                          (float)lr, (float)b1, (float)b2, (float)eps,
+                          (float)wd, (float)bc1, (float)bc2, has_master,
+                          dt_of(P), cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("ce_fwd", &ce_fwd, "fused logit cross-entropy fwd (loss + dlogits)");
+    m.def("add_relu_fwd", &add_relu_fwd);
+    m.def("add_relu_bwd", &add_relu_bwd);
+    m.def("bn_act_fwd", &bn_act_fwd);
+    m.def("bn_act_bwd", &bn_act_bwd);
+    m.def("sgd_step", &sgd_step);
+    m.def("adam_step", &adam_step);
+    m.attr("_built_for") = "gfx950";
+}

@@ -1,0 +1,349 @@
+// Fused BatchNorm(+residual add)(+ReLU), NHWC, training + eval, fwd + bwd.
+//
+// Replaces 3 separate launches (BN, add, ReLU) per ResNet block with one
+// stats pass + one apply pass (fwd) and one stats + one apply (bwd) — the
+// "fused BN+ReLU" requirement of BASELINE.json's north star. Reference
+// behavior: Flux BatchNorm in Metalhead blocks, with
+// per-replica running stats that are never synced across replicas.
+//
+// Layout: x is [rows = N*H*W][C] with C contiguous (torch channels_last).
+// Stats accumulate in fp32; bf16 IO is 16 B/lane vectorized.
+#include "fda_common.h"
+#include "fda_kernels.h"
+
+namespace fda {
+
+// block = 256 threads split into (rows-per-block) x (threads-per-row);
+// threads-per-row = chunkC / V so each thread owns V consecutive channels.
+template <typename T, int V>
+__global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
+                                float* __restrict__ sumsq, int64_t rows, int C,
+                                int c_base, int chunkC) {
+    extern __shared__ __attribute__((aligned(16))) float smem[];  // 256*V floats
+    const int tpr = chunkC / V;
+    const int rpb = blockDim.x / tpr;
+    const int lane_c = threadIdx.x % tpr;
+    const int sub_r = threadIdx.x / tpr;
+    const int c0 = c_base + lane_c * V;
+
+    float s[V], q[V];
+    #pragma unroll
+    for (int k = 0; k < V; ++k) s[k] = q[k] = 0.f;
+
+    for (int64_t r = (int64_t)blockIdx.x * rpb + sub_r; r < rows;
+         r += (int64_t)gridDim.x * rpb) {
+        T xv[V];
+        *(uint4*)xv = *(const uint4*)(x + r * C + c0);
+        #pragma unroll
+        for (int k = 0; k < V; ++k) {
+            float v = load_f32(xv + k);
+            s[k] += v;
+            q[k] += v * v;
+        }
+    }
+    // reduce over the rpb row-groups, one array at a time
+    #pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+        float* loc = pass == 0 ? s : q;
+        float* dst = pass == 0 ? sum : sumsq;
+        #pragma unroll
+        for (int k = 0; k < V; ++k) smem[threadIdx.x * V + k] = loc[k];
+        __syncthreads();
+        if (sub_r == 0) {
+            float acc[V];
+            #pragma unroll
+            for (int k = 0; k < V; ++k) acc[k] = 0.f;
+            for (int rr = 0; rr < rpb; ++rr) {
+                const float* src = smem + (rr * tpr + lane_c) * V;
+                #pragma unroll
+                for (int k = 0; k < V; ++k) acc[k] += src[k];
+            }
+            #pragma unroll
+            for (int k = 0; k < V; ++k) atomicAdd(dst + c0 + k, acc[k]);
+        }
+        __syncthreads();
+    }
+}
+
+__global__ void bn_finalize_kernel(float* __restrict__ ws,
+                                   const float* __restrict__ weight,
+                                   const float* __restrict__ bias,
+                                   float* __restrict__ rm, float* __restrict__ rv,
+                                   float* __restrict__ save_mean,
+                                   float* __restrict__ save_invstd, int64_t rows,
+                                   int C, int training, float momentum,
+                                   float eps) {
+    int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float mean, invstd;
+    if (training) {
+        const float inv_m = 1.f / (float)rows;
+        mean = ws[c] * inv_m;
+        float var = fmaxf(ws[C + c] * inv_m - mean * mean, 0.f);
+        invstd = rsqrtf(var + eps);
+        rm[c] += momentum * (mean - rm[c]);
+        const float unbias = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
+        rv[c] += momentum * (var * unbias - rv[c]);
+        save_mean[c] = mean;
+        save_invstd[c] = invstd;
+    } else {
+        mean = rm[c];
+        invstd = rsqrtf(rv[c] + eps);
+        save_mean[c] = mean;
+        save_invstd[c] = invstd;
+    }
+    const float scale = weight[c] * invstd;
+    ws[2 * C + c] = scale;
+    ws[3 * C + c] = bias[c] - mean * scale;
+}
+
+template <typename T, int V, bool RELU, bool RES>
+__global__ void bn_apply_kernel(const T* __restrict__ x,
+                                const T* __restrict__ res, T* __restrict__ out,
+                                const float* __restrict__ scale,
+                                const float* __restrict__ shift, int64_t nvec,
+                                int C) {
+    const int cvec = C / V;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const int c0 = (int)(i % cvec) * V;
+        T xv[V], rv_[V], ov[V];
+        *(uint4*)xv = ((const uint4*)x)[i];
+        if constexpr (RES) *(uint4*)rv_ = ((const uint4*)res)[i];
+        #pragma unroll
+        for (int k = 0; k < V; ++k) {
+            float v = load_f32(xv + k) * scale[c0 + k] + shift[c0 + k];
+            if constexpr (RES) v += load_f32(rv_ + k);
+            if constexpr (RELU) v = fmaxf(v, 0.f);
+            store_f32(ov + k, v);
+        }
+        ((uint4*)out)[i] = *(uint4*)ov;
+    }
+}
+
+// ---- backward -------------------------------------------------------------
+
+template <typename T, int V, bool RELU>
+__global__ void bn_bwd_stats_kernel(const T* __restrict__ gout,
+                                    const T* __restrict__ x,
+                                    const T* __restrict__ out,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    float* __restrict__ sum_g,
+                                    float* __restrict__ sum_gx, int64_t rows,
+                                    int C, int c_base, int chunkC) {
+    extern __shared__ __attribute__((aligned(16))) float smem[];
+    const int tpr = chunkC / V;
+    const int rpb = blockDim.x / tpr;
+    const int lane_c = threadIdx.x % tpr;
+    const int sub_r = threadIdx.x / tpr;
+    const int c0 = c_base + lane_c * V;
+
+    float mu[V], is[V], sg[V], sgx[V];
+    #pragma unroll
+    for (int k = 0; k < V; ++k) {
+        mu[k] = mean[c0 + k];
+        is[k] = invstd[c0 + k];
+        sg[k] = sgx[k] = 0.f;
+    }
+    for (int64_t r = (int64_t)blockIdx.x * rpb + sub_r; r < rows;
+         r += (int64_t)gridDim.x * rpb) {
+        T gv[V], xv[V], ov[V];
+        *(uint4*)gv = *(const uint4*)(gout + r * C + c0);
+        *(uint4*)xv = *(const uint4*)(x + r * C + c0);
+        if constexpr (RELU) *(uint4*)ov = *(const uint4*)(out + r * C + c0);
+        #pragma unroll
+        for (int k = 0; k < V; ++k) {
+            float g = load_f32(gv + k);
+            if constexpr (RELU) g = load_f32(ov + k) > 0.f ? g : 0.f;
+            const float xhat = (load_f32(xv + k) - mu[k]) * is[k];
+            sg[k] += g;
+            sgx[k] += g * xhat;
+        }
+    }
+    #pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+        float* loc = pass == 0 ? sg : sgx;
+        float* dst = pass == 0 ? sum_g : sum_gx;
+        #pragma unroll
+        for (int k = 0; k < V; ++k) smem[threadIdx.x * V + k] = loc[k];
+        __syncthreads();
+        if (sub_r == 0) {
+            float acc[V];
+            #pragma unroll
+            for (int k = 0; k < V; ++k) acc[k] = 0.f;
+            for (int rr = 0; rr < rpb; ++rr) {
+                const float* src = smem + (rr * tpr + lane_c) * V;
+                #pragma unroll
+                for (int k = 0; k < V; ++k) acc[k] += src[k];
+            }
+            #pragma unroll
+            for (int k = 0; k < V; ++k) atomicAdd(dst + c0 + k, acc[k]);
+        }
+        __syncthreads();
+    }
+}
+
+__global__ void bn_bwd_finalize_kernel(float* __restrict__ ws,
+                                       float* __restrict__ gw,
+                                       float* __restrict__ gb, int64_t rows,
+                                       int C, int training) {
+    int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    const float sg = ws[c], sgx = ws[C + c];
+    gb[c] = sg;
+    gw[c] = sgx;
+    const float inv_m = training ? 1.f / (float)rows : 0.f;
+    ws[2 * C + c] = sg * inv_m;   // k1
+    ws[3 * C + c] = sgx * inv_m;  // k2 (0 in eval mode -> gx = scale*g)
+}
+
+template <typename T, int V, bool RELU>
+__global__ void bn_bwd_apply_kernel(const T* __restrict__ gout,
+                                    const T* __restrict__ x,
+                                    const T* __restrict__ out,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    const float* __restrict__ weight,
+                                    const float* __restrict__ k1,
+                                    const float* __restrict__ k2,
+                                    T* __restrict__ gx, int64_t nvec, int C) {
+    const int cvec = C / V;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const int c0 = (int)(i % cvec) * V;
+        T gv[V], xv[V], ov[V], rv_[V];
+        *(uint4*)gv = ((const uint4*)gout)[i];
+        *(uint4*)xv = ((const uint4*)x)[i];
+        if constexpr (RELU) *(uint4*)ov = ((const uint4*)out)[i];
+        #pragma unroll
+        for (int k = 0; k < V; ++k) {
+            const int c = c0 + k;
+            float g = load_f32(gv + k);
+            if constexpr (RELU) g = load_f32(ov + k) > 0.f ? g : 0.f;
+            const float is = invstd[c];
+            const float xhat = (load_f32(xv + k) - mean[c]) * is;
+            const float r = (g - k1[c] - xhat * k2[c]) * weight[c] * is;
+            store_f32(rv_ + k, r);
+        }
+        ((uint4*)gx)[i] = *(uint4*)rv_;
+    }
+}
+
+// ---- launchers ------------------------------------------------------------
+
+static inline void stats_geom(int C, int V, int64_t rows, int& chunkC,
+                              int& nchunks, int& grid, int& shmem) {
+    const int maxC = 256 * V;
+    chunkC = C < maxC ? C : maxC;
+    nchunks = (C + chunkC - 1) / chunkC;
+    const int tpr = chunkC / V;
+    const int rpb = 256 / tpr;
+    int64_t blocks = (rows + rpb - 1) / rpb;
+    grid = (int)(blocks < 1 ? 1 : (blocks > 2048 ? 2048 : blocks));
+    shmem = 256 * V * (int)sizeof(float);
+}
+
+void bn_stats_launch(const void* x, float* ws, int64_t rows, int C, DT dt,
+                     hipStream_t s) {
+    const int V = dt == DT::BF16 ? 8 : 4;
+    int chunkC, nchunks, grid, shmem;
+    stats_geom(C, V, rows, chunkC, nchunks, grid, shmem);
+    for (int ch = 0; ch < nchunks; ++ch) {
+        const int c_base = ch * chunkC;
+        const int cc = (C - c_base) < chunkC ? (C - c_base) : chunkC;
+        if (dt == DT::BF16)
+            hipLaunchKernelGGL((bn_stats_kernel<unsigned short, 8>), dim3(grid),
+                               dim3(256), shmem, s, (const unsigned short*)x,
+                               ws, ws + C, rows, C, c_base, cc);
+        else
+            hipLaunchKernelGGL((bn_stats_kernel<float, 4>), dim3(grid),
+                               dim3(256), shmem, s, (const float*)x, ws, ws + C,
+                               rows, C, c_base, cc);
+    }
+}
+
+void bn_finalize_launch(float* ws, const float* weight, const float* bias,
+                        float* running_mean, float* running_var,
+                        float* save_mean, float* save_invstd, int64_t rows,
+                        int C, bool training, float momentum, float eps,
+                        hipStream_t s) {
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                       s, ws, weight, bias, running_mean, running_var,
+                       save_mean, save_invstd, rows, C, training ? 1 : 0,
+                       momentum, eps);
+}
+
+void bn_apply_launch(const void* x, const void* residual, void* out,
+                     const float* ws, int64_t rows, int C, bool relu, DT dt,
+                     hipStream_t s) {
+    const int V = dt == DT::BF16 ? 8 : 4;
+    const int64_t nvec = rows * C / V;
+    int64_t blocks = (nvec + 255) / 256;
+    const int grid = (int)(blocks < 1 ? 1 : (blocks > 4096 ? 4096 : blocks));
+    const float* scale = ws + 2 * C;
+    const float* shift = ws + 3 * C;
+    #define FDA_APPLY(T, VW, RELU_, RES_)                                       \
+        hipLaunchKernelGGL((bn_apply_kernel<T, VW, RELU_, RES_>), dim3(grid),   \
+                           dim3(256), 0, s, (const T*)x, (const T*)residual,    \
+                           (T*)out, scale, shift, nvec, C)
+    if (dt == DT::BF16) {
+        if (relu) { if (residual) FDA_APPLY(unsigned short, 8, true, true); else FDA_APPLY(unsigned short, 8, true, false); }
+        else      { if (residual) FDA_APPLY(unsigned short, 8, false, true); else FDA_APPLY(unsigned short, 8, false, false); }
+    } else {
+        if (relu) { if (residual) FDA_APPLY(float, 4, true, true); else FDA_APPLY(float, 4, true, false); }
+        else      { if (residual) FDA_APPLY(float, 4, false, true); else FDA_APPLY(float, 4, false, false); }
+    }
+    #undef FDA_APPLY
+}
+
+void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
+                         const float* save_mean, const float* save_invstd,
+                         float* ws, int64_t rows, int C, bool relu, DT dt,
+                         hipStream_t s) {
+    const int V = dt == DT::BF16 ? 8 : 4;
+    int chunkC, nchunks, grid, shmem;
+    stats_geom(C, V, rows, chunkC, nchunks, grid, shmem);
+    for (int ch = 0; ch < nchunks; ++ch) {
+        const int c_base = ch * chunkC;
+        const int cc = (C - c_base) < chunkC ? (C - c_base) : chunkC;
+        #define FDA_BSTATS(T, VW, RELU_)                                        \
+            hipLaunchKernelGGL((bn_bwd_stats_kernel<T, VW, RELU_>), dim3(grid), \
+                               dim3(256), shmem, s, (const T*)gout,             \
+                               (const T*)x, (const T*)out, save_mean,           \
+                               save_invstd, ws, ws + C, rows, C, c_base, cc)
+        if (dt == DT::BF16) { if (relu) FDA_BSTATS(unsigned short, 8, true); else FDA_BSTATS(unsigned short, 8, false); }
+        else { if (relu) FDA_BSTATS(float, 4, true); else FDA_BSTATS(float, 4, false); }
+        #undef FDA_BSTATS
+    }
+}
+
+void bn_bwd_finalize_launch(float* ws, const float* /*weight*/,
+                            const float* /*save_invstd*/, float* gw, float* gb,
+                            int64_t rows, int C, bool training, hipStream_t s) {
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                       0, s, ws, gw, gb, rows, C, training ? 1 : 0);
+}
+
+void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
+                         const float* save_mean, const float* save_invstd,
+                         const float* weight, const float* ws, void* gx,
+                         int64_t rows, int C, bool relu, bool /*training*/,
+                         DT dt, hipStream_t s) {
+    const int V = dt == DT::BF16 ? 8 : 4;
+    const int64_t nvec = rows * C / V;
+    int64_t blocks = (nvec + 255) / 256;
+    const int grid = (int)(blocks < 1 ? 1 : (blocks > 4096 ? 4096 : blocks));
+    const float* k1 = ws + 2 * C;
+    const float* k2 = ws + 3 * C;
+    #define FDA_BAPPLY(T, VW, RELU_)                                            \
+        hipLaunchKernelGGL((bn_bwd_apply_kernel<T, VW, RELU_>), dim3(grid),     \
+                           dim3(256), 0, s, (const T*)gout, (const T*)x,        \
+                           (const T*)out, save_mean, save_invstd, weight, k1,   \
+                           k2, (T*)gx, nvec, C)
+    if (dt == DT::BF16) { if (relu) FDA_BAPPLY(unsigned short, 8, true); else FDA_BAPPLY(unsigned short, 8, false); }
+    else { if (relu) FDA_BAPPLY(float, 4, true); else FDA_BAPPLY(float, 4, false); }
+    #undef FDA_BAPPLY
+}
+
+}  // namespace fda

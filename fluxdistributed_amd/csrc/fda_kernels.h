@@ -1,0 +1,61 @@
+// Launcher API for the gfx950 kernels. bf16 tensors cross this boundary as
+// raw unsigned short bit patterns (at::BFloat16 is bit-compatible).
+#pragma once
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+namespace fda {
+
+enum class DT { F32 = 0, BF16 = 1 };
+
+// fused logit cross-entropy: mean loss over N rows + dlogits in one pass
+void ce_fwd_launch(const void* logits, const int64_t* target, float* loss,
+                   void* dlogits, int N, int C, DT dt, hipStream_t s);
+
+// out = relu(x + r)
+void add_relu_fwd_launch(const void* x, const void* r, void* out, int64_t n,
+                         DT dt, hipStream_t s);
+// gx = gout * (out > 0)
+void add_relu_bwd_launch(const void* gout, const void* out, void* gx, int64_t n,
+                         DT dt, hipStream_t s);
+
+// BatchNorm(+residual)(+ReLU), NHWC. ws layout (floats):
+//   [0,C)    sum        [C,2C)  sumsq
+//   [2C,3C)  scale      [3C,4C) shift
+// save_mean/save_invstd are separate C-float buffers.
+void bn_stats_launch(const void* x, float* ws, int64_t rows, int C, DT dt,
+                     hipStream_t s);
+void bn_finalize_launch(float* ws, const float* weight, const float* bias,
+                        float* running_mean, float* running_var,
+                        float* save_mean, float* save_invstd, int64_t rows,
+                        int C, bool training, float momentum, float eps,
+                        hipStream_t s);
+void bn_apply_launch(const void* x, const void* residual, void* out,
+                     const float* ws /*scale/shift*/, int64_t rows, int C,
+                     bool relu, DT dt, hipStream_t s);
+
+// backward. ws layout (floats): [0,C) sum_g  [C,2C) sum_g_xhat
+//   [2C,3C) k1  [3C,4C) k2   (k's folded with invstd*gamma in finalize)
+void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
+                         const float* save_mean, const float* save_invstd,
+                         float* ws, int64_t rows, int C, bool relu, DT dt,
+                         hipStream_t s);
+void bn_bwd_finalize_launch(float* ws, const float* weight,
+                            const float* save_invstd, float* gw, float* gb,
+                            int64_t rows, int C, bool training, hipStream_t s);
+void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
+                         const float* save_mean, const float* save_invstd,
+                         const float* weight, const float* ws, void* gx,
+                         int64_t rows, int C, bool relu, bool training, DT dt,
+                         hipStream_t s);
+
+// flat fused optimizers. P: param dtype; M/V/S fp32; G param dtype.
+void sgd_step_launch(void* P, const void* G, float* M, float* V, int64_t n,
+                     float lr, float mom, float wd, bool nesterov,
+                     bool has_master, DT dt, hipStream_t s);
+void adam_step_launch(void* P, const void* G, float* M, float* V, float* S,
+                      int64_t n, float lr, float b1, float b2, float eps,
+                      float wd, float bc1, float bc2, bool has_master, DT dt,
+                      hipStream_t s);
+
+}  // namespace fda

@@ -1,0 +1,176 @@
+"""Fused training ops: CPU reference path + gfx950 native path.
+
+Each op exists because it is on the ResNet hot path of the reference
+(/root/reference/src/ddp_tasks.jl:28 loss; Metalhead basic blocks BN+ReLU;
+residual adds) — see SURVEY.md §2.4 for the full kernel inventory.
+
+Numerics contract (tests/test_ops_gpu.py): every native kernel is compared
+against the plain fp32 PyTorch composition of the same op.
+"""
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from .native import require_native
+
+
+def _on_gpu(*tensors) -> bool:
+    return any(t is not None and isinstance(t, torch.Tensor) and t.is_cuda for t in tensors)
+
+
+# --------------------------------------------------------------------------
+# Fused logit cross-entropy (reference: Flux.Losses.logitcrossentropy,
+# /root/reference/src/ddp_tasks.jl:28). Forward computes the mean loss AND
+# d(loss)/d(logits) in one read of the logits; backward just scales.
+# --------------------------------------------------------------------------
+
+
+class _LogitCrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits: torch.Tensor, target: torch.Tensor):
+        # target: int64 class indices, shape (N,)
+        if logits.is_cuda:
+            C = require_native("logit_cross_entropy")
+            loss, dlogits = C.ce_fwd(logits, target)
+        else:
+            x = logits.float()
+            lse = torch.logsumexp(x, dim=1, keepdim=True)
+            logp = x - lse
+            n = logits.shape[0]
+            loss = -logp[torch.arange(n), target].mean()
+            sm = torch.exp(logp)
+            sm[torch.arange(n), target] -= 1.0
+            dlogits = (sm / n).to(logits.dtype)
+            loss = loss.to(logits.dtype)
+        ctx.save_for_backward(dlogits)
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (dlogits,) = ctx.saved_tensors
+        return dlogits * grad_out, None
+
+
+def logit_cross_entropy(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """Mean cross-entropy over un-normalized logits; target = class indices."""
+    return _LogitCrossEntropy.apply(logits, target.long())
+
+
+# --------------------------------------------------------------------------
+# Fused residual add + ReLU (tail of every ResNet block).
+# --------------------------------------------------------------------------
+
+
+class _AddReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, residual: torch.Tensor):
+        if x.is_cuda:
+            C = require_native("fused_add_relu")
+            out = C.add_relu_fwd(x, residual)
+        else:
+            out = torch.relu(x + residual)
+        ctx.save_for_backward(out)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (out,) = ctx.saved_tensors
+        if grad_out.is_cuda:
+            C = require_native("fused_add_relu")
+            gx = C.add_relu_bwd(grad_out.contiguous(memory_format=torch.channels_last)
+                                if grad_out.dim() == 4 else grad_out.contiguous(), out)
+        else:
+            gx = grad_out * (out > 0).to(grad_out.dtype)
+        return gx, gx
+
+
+def fused_add_relu(x: torch.Tensor, residual: torch.Tensor) -> torch.Tensor:
+    return _AddReLU.apply(x, residual)
+
+
+# --------------------------------------------------------------------------
+# Fused BatchNorm (+ optional residual add) + ReLU, NHWC, training & eval.
+# Matches the reference semantic (SURVEY.md §7 hard-part 3): per-replica
+# running stats, never synced across data-parallel replicas.
+# --------------------------------------------------------------------------
+
+
+class _BNAct(torch.autograd.Function):
+    @staticmethod
+    def forward(
+        ctx,
+        x: torch.Tensor,
+        weight: torch.Tensor,
+        bias: torch.Tensor,
+        running_mean: torch.Tensor,
+        running_var: torch.Tensor,
+        training: bool,
+        momentum: float,
+        eps: float,
+        relu: bool,
+        residual: Optional[torch.Tensor],
+    ):
+        C = require_native("batch_norm_act")
+        out, save_mean, save_invstd = C.bn_act_fwd(
+            x, weight, bias, running_mean, running_var,
+            training, momentum, eps, relu,
+            residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype),
+        )
+        ctx.save_for_backward(x, weight, save_mean, save_invstd, out)
+        ctx.relu = relu
+        ctx.has_residual = residual is not None
+        ctx.training = training
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        x, weight, save_mean, save_invstd, out = ctx.saved_tensors
+        C = require_native("batch_norm_act")
+        gx, gw, gb = C.bn_act_bwd(
+            grad_out.contiguous(memory_format=torch.channels_last),
+            x, weight, save_mean, save_invstd, out, ctx.relu, ctx.training,
+        )
+        gres = gx_res = None
+        if ctx.has_residual:
+            # d(out)/d(residual) = relu-mask * grad_out = first stage of gx;
+            # the kernel returns it as gx's sibling (same mask, no bn chain).
+            gx_res = C.add_relu_bwd(grad_out.contiguous(memory_format=torch.channels_last), out) \
+                if ctx.relu else grad_out
+            gres = gx_res
+        return gx, gw, gb, None, None, None, None, None, None, gres
+
+
+def batch_norm_act(
+    x: torch.Tensor,
+    weight: torch.Tensor,
+    bias: torch.Tensor,
+    running_mean: torch.Tensor,
+    running_var: torch.Tensor,
+    training: bool,
+    momentum: float = 0.1,
+    eps: float = 1e-5,
+    relu: bool = True,
+    residual: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """BN (train or eval stats) + optional residual add + optional ReLU.
+
+    GPU: one fused native kernel pair (NHWC). CPU: composed PyTorch ops —
+    identical math, used as the oracle.
+    """
+    if _on_gpu(x):
+        return _BNAct.apply(
+            x, weight, bias, running_mean, running_var,
+            training, momentum, eps, relu, residual,
+        )
+    # CPU reference path: compute in fp32 (handles mixed bf16-x/fp32-params)
+    out = F.batch_norm(
+        x.float(), running_mean, running_var, weight.float(), bias.float(),
+        training, momentum, eps,
+    )
+    if residual is not None:
+        out = out + residual.float()
+    if relu:
+        out = torch.relu(out)
+    return out.to(x.dtype)
