@@ -93,9 +93,11 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
     auto stream = cur_stream();
 
     if (training) {
-        ws.narrow(0, 0, 2 * C).zero_();
-        fda::bn_stats_launch(x.data_ptr(), ws.data_ptr<float>(), rows, (int)C,
-                             dt_of(x), stream);
+        auto part = at::empty({fda::bn_stats_partial_floats((int)C, rows, dt_of(x))},
+                              fopts);
+        fda::bn_stats_launch(x.data_ptr(), ws.data_ptr<float>(),
+                             part.data_ptr<float>(), rows, (int)C, dt_of(x),
+                             stream);
     }
     fda::bn_finalize_launch(ws.data_ptr<float>(), weight.data_ptr<float>(),
                             bias.data_ptr<float>(),
@@ -116,16 +118,19 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
     auto [rows, C] = nhwc_rows(x);
     auto gc = gout.contiguous(at::MemoryFormat::ChannelsLast);
     auto fopts = x.options().dtype(at::kFloat);
-    auto ws = at::zeros({4 * C}, fopts);
+    auto ws = at::empty({4 * C}, fopts);
     auto gw = at::empty({C}, fopts);
     auto gb = at::empty({C}, fopts);
     auto gx = at::empty_like(x);
     auto stream = cur_stream();
 
+    auto part = at::empty({fda::bn_stats_partial_floats((int)C, rows, dt_of(x))},
+                          fopts);
     fda::bn_bwd_stats_launch(gc.data_ptr(), x.data_ptr(), out.data_ptr(),
                              save_mean.data_ptr<float>(),
                              save_invstd.data_ptr<float>(), ws.data_ptr<float>(),
-                             rows, (int)C, relu, dt_of(x), stream);
+                             part.data_ptr<float>(), rows, (int)C, relu,
+                             dt_of(x), stream);
     fda::bn_bwd_finalize_launch(ws.data_ptr<float>(), weight.data_ptr<float>(),
                                 save_invstd.data_ptr<float>(),
                                 gw.data_ptr<float>(), gb.data_ptr<float>(),

@@ -15,9 +15,12 @@ namespace fda {
 
 // block = 256 threads split into (rows-per-block) x (threads-per-row);
 // threads-per-row = chunkC / V so each thread owns V consecutive channels.
+// Each block writes PARTIAL per-channel sums to part[block][2*chunkC]
+// (plain coalesced stores — a global atomicAdd per channel serializes on
+// L2 and was measured 30x slower); bn_stats_reduce_kernel folds partials.
 template <typename T, int V>
-__global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
-                                float* __restrict__ sumsq, int64_t rows, int C,
+__global__ void bn_stats_kernel(const T* __restrict__ x,
+                                float* __restrict__ part, int64_t rows, int C,
                                 int c_base, int chunkC) {
     extern __shared__ __attribute__((aligned(16))) float smem[];  // 256*V floats
     const int tpr = chunkC / V;
@@ -30,22 +33,34 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum
     #pragma unroll
     for (int k = 0; k < V; ++k) s[k] = q[k] = 0.f;
 
+    // 2 independent row streams per iteration for memory-level parallelism
+    const int64_t stride = (int64_t)gridDim.x * rpb;
     for (int64_t r = (int64_t)blockIdx.x * rpb + sub_r; r < rows;
-         r += (int64_t)gridDim.x * rpb) {
-        T xv[V];
+         r += 2 * stride) {
+        T xv[V], xw[V];
         *(uint4*)xv = *(const uint4*)(x + r * C + c0);
+        const bool second = r + stride < rows;
+        if (second) *(uint4*)xw = *(const uint4*)(x + (r + stride) * C + c0);
         #pragma unroll
         for (int k = 0; k < V; ++k) {
             float v = load_f32(xv + k);
             s[k] += v;
             q[k] += v * v;
         }
+        if (second) {
+            #pragma unroll
+            for (int k = 0; k < V; ++k) {
+                float v = load_f32(xw + k);
+                s[k] += v;
+                q[k] += v * v;
+            }
+        }
     }
     // reduce over the rpb row-groups, one array at a time
+    float* out = part + (int64_t)blockIdx.x * 2 * chunkC;
     #pragma unroll
     for (int pass = 0; pass < 2; ++pass) {
         float* loc = pass == 0 ? s : q;
-        float* dst = pass == 0 ? sum : sumsq;
         #pragma unroll
         for (int k = 0; k < V; ++k) smem[threadIdx.x * V + k] = loc[k];
         __syncthreads();
@@ -59,9 +74,45 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum
                 for (int k = 0; k < V; ++k) acc[k] += src[k];
             }
             #pragma unroll
-            for (int k = 0; k < V; ++k) atomicAdd(dst + c0 + k, acc[k]);
+            for (int k = 0; k < V; ++k)
+                out[pass * chunkC + lane_c * V + k] = acc[k];
         }
         __syncthreads();
+    }
+}
+
+// Fold part[NB][2*chunkC] -> sum/sumsq[chunkC]. Block: cpb channel lanes x
+// nsub partial-subgroups; LDS-reduce over subgroups.
+__global__ void bn_stats_reduce_kernel(const float* __restrict__ part,
+                                       float* __restrict__ out_s,
+                                       float* __restrict__ out_q, int NB,
+                                       int chunkC, int c_base) {
+    __shared__ float sm[2 * 256];
+    const int cpb = chunkC < (int)blockDim.x ? chunkC : (int)blockDim.x;
+    const int nsub = blockDim.x / cpb;
+    const int lane = threadIdx.x % cpb;
+    const int sub = threadIdx.x / cpb;
+    const int cc = blockIdx.x * cpb + lane;
+    if (cc >= chunkC) return;
+    float s = 0.f, q = 0.f;
+    for (int b = sub; b < NB; b += nsub) {
+        s += part[(int64_t)b * 2 * chunkC + cc];
+        q += part[(int64_t)b * 2 * chunkC + chunkC + cc];
+    }
+    if (nsub > 1) {
+        sm[threadIdx.x] = s;
+        sm[256 + threadIdx.x] = q;
+        __syncthreads();
+        if (sub == 0) {
+            for (int t = 1; t < nsub; ++t) {
+                s += sm[t * cpb + lane];
+                q += sm[256 + t * cpb + lane];
+            }
+        }
+    }
+    if (sub == 0) {
+        out_s[c_base + cc] = s;
+        out_q[c_base + cc] = q;
     }
 }
 
@@ -129,8 +180,7 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ gout,
                                     const T* __restrict__ out,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
-                                    float* __restrict__ sum_g,
-                                    float* __restrict__ sum_gx, int64_t rows,
+                                    float* __restrict__ part, int64_t rows,
                                     int C, int c_base, int chunkC) {
     extern __shared__ __attribute__((aligned(16))) float smem[];
     const int tpr = chunkC / V;
@@ -161,10 +211,10 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ gout,
             sgx[k] += g * xhat;
         }
     }
+    float* outp = part + (int64_t)blockIdx.x * 2 * chunkC;
     #pragma unroll
     for (int pass = 0; pass < 2; ++pass) {
         float* loc = pass == 0 ? sg : sgx;
-        float* dst = pass == 0 ? sum_g : sum_gx;
         #pragma unroll
         for (int k = 0; k < V; ++k) smem[threadIdx.x * V + k] = loc[k];
         __syncthreads();
@@ -178,7 +228,8 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ gout,
                 for (int k = 0; k < V; ++k) acc[k] += src[k];
             }
             #pragma unroll
-            for (int k = 0; k < V; ++k) atomicAdd(dst + c0 + k, acc[k]);
+            for (int k = 0; k < V; ++k)
+                outp[pass * chunkC + lane_c * V + k] = acc[k];
         }
         __syncthreads();
     }
@@ -240,12 +291,19 @@ static inline void stats_geom(int C, int V, int64_t rows, int& chunkC,
     const int tpr = chunkC / V;
     const int rpb = 256 / tpr;
     int64_t blocks = (rows + rpb - 1) / rpb;
-    grid = (int)(blocks < 1 ? 1 : (blocks > 2048 ? 2048 : blocks));
+    grid = (int)(blocks < 1 ? 1 : (blocks > 512 ? 512 : blocks));
     shmem = 256 * V * (int)sizeof(float);
 }
 
-void bn_stats_launch(const void* x, float* ws, int64_t rows, int C, DT dt,
-                     hipStream_t s) {
+int bn_stats_partial_floats(int C, int64_t rows, DT dt) {
+    const int V = dt == DT::BF16 ? 8 : 4;
+    int chunkC, nchunks, grid, shmem;
+    stats_geom(C, V, rows, chunkC, nchunks, grid, shmem);
+    return grid * 2 * chunkC;
+}
+
+void bn_stats_launch(const void* x, float* ws, float* part, int64_t rows,
+                     int C, DT dt, hipStream_t s) {
     const int V = dt == DT::BF16 ? 8 : 4;
     int chunkC, nchunks, grid, shmem;
     stats_geom(C, V, rows, chunkC, nchunks, grid, shmem);
@@ -255,11 +313,15 @@ void bn_stats_launch(const void* x, float* ws, int64_t rows, int C, DT dt,
         if (dt == DT::BF16)
             hipLaunchKernelGGL((bn_stats_kernel<unsigned short, 8>), dim3(grid),
                                dim3(256), shmem, s, (const unsigned short*)x,
-                               ws, ws + C, rows, C, c_base, cc);
+                               part, rows, C, c_base, cc);
         else
             hipLaunchKernelGGL((bn_stats_kernel<float, 4>), dim3(grid),
-                               dim3(256), shmem, s, (const float*)x, ws, ws + C,
+                               dim3(256), shmem, s, (const float*)x, part,
                                rows, C, c_base, cc);
+        const int cpb = cc < 256 ? cc : 256;
+        hipLaunchKernelGGL(bn_stats_reduce_kernel,
+                           dim3((cc + cpb - 1) / cpb), dim3(256), 0, s, part,
+                           ws, ws + C, grid, cc, c_base);
     }
 }
 
@@ -299,8 +361,8 @@ void bn_apply_launch(const void* x, const void* residual, void* out,
 
 void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
                          const float* save_mean, const float* save_invstd,
-                         float* ws, int64_t rows, int C, bool relu, DT dt,
-                         hipStream_t s) {
+                         float* ws, float* part, int64_t rows, int C,
+                         bool relu, DT dt, hipStream_t s) {
     const int V = dt == DT::BF16 ? 8 : 4;
     int chunkC, nchunks, grid, shmem;
     stats_geom(C, V, rows, chunkC, nchunks, grid, shmem);
@@ -311,10 +373,14 @@ void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
             hipLaunchKernelGGL((bn_bwd_stats_kernel<T, VW, RELU_>), dim3(grid), \
                                dim3(256), shmem, s, (const T*)gout,             \
                                (const T*)x, (const T*)out, save_mean,           \
-                               save_invstd, ws, ws + C, rows, C, c_base, cc)
+                               save_invstd, part, rows, C, c_base, cc)
         if (dt == DT::BF16) { if (relu) FDA_BSTATS(unsigned short, 8, true); else FDA_BSTATS(unsigned short, 8, false); }
         else { if (relu) FDA_BSTATS(float, 4, true); else FDA_BSTATS(float, 4, false); }
         #undef FDA_BSTATS
+        const int cpb = cc < 256 ? cc : 256;
+        hipLaunchKernelGGL(bn_stats_reduce_kernel,
+                           dim3((cc + cpb - 1) / cpb), dim3(256), 0, s, part,
+                           ws, ws + C, grid, cc, c_base);
     }
 }
 

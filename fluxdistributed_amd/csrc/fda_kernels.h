@@ -23,8 +23,9 @@ void add_relu_bwd_launch(const void* gout, const void* out, void* gx, int64_t n,
 //   [0,C)    sum        [C,2C)  sumsq
 //   [2C,3C)  scale      [3C,4C) shift
 // save_mean/save_invstd are separate C-float buffers.
-void bn_stats_launch(const void* x, float* ws, int64_t rows, int C, DT dt,
-                     hipStream_t s);
+int bn_stats_partial_floats(int C, int64_t rows, DT dt);
+void bn_stats_launch(const void* x, float* ws, float* part, int64_t rows,
+                     int C, DT dt, hipStream_t s);
 void bn_finalize_launch(float* ws, const float* weight, const float* bias,
                         float* running_mean, float* running_var,
                         float* save_mean, float* save_invstd, int64_t rows,
@@ -38,8 +39,8 @@ void bn_apply_launch(const void* x, const void* residual, void* out,
 //   [2C,3C) k1  [3C,4C) k2   (k's folded with invstd*gamma in finalize)
 void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
                          const float* save_mean, const float* save_invstd,
-                         float* ws, int64_t rows, int C, bool relu, DT dt,
-                         hipStream_t s);
+                         float* ws, float* part, int64_t rows, int C,
+                         bool relu, DT dt, hipStream_t s);
 void bn_bwd_finalize_launch(float* ws, const float* weight,
                             const float* save_invstd, float* gw, float* gb,
                             int64_t rows, int C, bool training, hipStream_t s);
