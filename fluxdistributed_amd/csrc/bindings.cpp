@@ -93,19 +93,26 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
     auto stream = cur_stream();
 
     if (training) {
+        TORCH_CHECK(C % 64 == 0, "training BN needs C % 64 == 0, got ", C);
         auto part = at::empty({fda::bn_stats_partial_floats((int)C, rows, dt_of(x))},
                               fopts);
         fda::bn_stats_launch(x.data_ptr(), ws.data_ptr<float>(),
-                             part.data_ptr<float>(), rows, (int)C, dt_of(x),
-                             stream);
+                             part.data_ptr<float>(), weight.data_ptr<float>(),
+                             bias.data_ptr<float>(),
+                             running_mean.data_ptr<float>(),
+                             running_var.data_ptr<float>(),
+                             save_mean.data_ptr<float>(),
+                             save_invstd.data_ptr<float>(), rows, (int)C,
+                             (float)momentum, (float)eps, dt_of(x), stream);
+    } else {
+        fda::bn_finalize_launch(ws.data_ptr<float>(), weight.data_ptr<float>(),
+                                bias.data_ptr<float>(),
+                                running_mean.data_ptr<float>(),
+                                running_var.data_ptr<float>(),
+                                save_mean.data_ptr<float>(),
+                                save_invstd.data_ptr<float>(), rows, (int)C,
+                                training, (float)momentum, (float)eps, stream);
     }
-    fda::bn_finalize_launch(ws.data_ptr<float>(), weight.data_ptr<float>(),
-                            bias.data_ptr<float>(),
-                            running_mean.data_ptr<float>(),
-                            running_var.data_ptr<float>(),
-                            save_mean.data_ptr<float>(),
-                            save_invstd.data_ptr<float>(), rows, (int)C,
-                            training, (float)momentum, (float)eps, stream);
     fda::bn_apply_launch(x.data_ptr(), has_res ? resc.data_ptr() : nullptr,
                          out.data_ptr(), ws.data_ptr<float>(), rows, (int)C,
                          relu, dt_of(x), stream);
@@ -124,17 +131,15 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
     auto gx = at::empty_like(x);
     auto stream = cur_stream();
 
+    TORCH_CHECK(C % 64 == 0, "BN bwd needs C % 64 == 0, got ", C);
     auto part = at::empty({fda::bn_stats_partial_floats((int)C, rows, dt_of(x))},
                           fopts);
     fda::bn_bwd_stats_launch(gc.data_ptr(), x.data_ptr(), out.data_ptr(),
                              save_mean.data_ptr<float>(),
                              save_invstd.data_ptr<float>(), ws.data_ptr<float>(),
-                             part.data_ptr<float>(), rows, (int)C, relu,
-                             dt_of(x), stream);
-    fda::bn_bwd_finalize_launch(ws.data_ptr<float>(), weight.data_ptr<float>(),
-                                save_invstd.data_ptr<float>(),
-                                gw.data_ptr<float>(), gb.data_ptr<float>(),
-                                rows, (int)C, training, stream);
+                             part.data_ptr<float>(), gw.data_ptr<float>(),
+                             gb.data_ptr<float>(), rows, (int)C, relu,
+                             training, dt_of(x), stream);
     fda::bn_bwd_apply_launch(gc.data_ptr(), x.data_ptr(), out.data_ptr(),
                              save_mean.data_ptr<float>(),
                              save_invstd.data_ptr<float>(),

@@ -24,8 +24,14 @@ void add_relu_bwd_launch(const void* gout, const void* out, void* gx, int64_t n,
 //   [2C,3C)  scale      [3C,4C) shift
 // save_mean/save_invstd are separate C-float buffers.
 int bn_stats_partial_floats(int C, int64_t rows, DT dt);
-void bn_stats_launch(const void* x, float* ws, float* part, int64_t rows,
-                     int C, DT dt, hipStream_t s);
+// training: stats partials + fused reduce/finalize (running-stat update,
+// scale/shift into ws[2C..4C))
+void bn_stats_launch(const void* x, float* ws, float* part,
+                     const float* weight, const float* bias,
+                     float* running_mean, float* running_var, float* save_mean,
+                     float* save_invstd, int64_t rows, int C, float momentum,
+                     float eps, DT dt, hipStream_t s);
+// eval: scale/shift from running stats
 void bn_finalize_launch(float* ws, const float* weight, const float* bias,
                         float* running_mean, float* running_var,
                         float* save_mean, float* save_invstd, int64_t rows,
@@ -37,13 +43,12 @@ void bn_apply_launch(const void* x, const void* residual, void* out,
 
 // backward. ws layout (floats): [0,C) sum_g  [C,2C) sum_g_xhat
 //   [2C,3C) k1  [3C,4C) k2   (k's folded with invstd*gamma in finalize)
+// bwd: stats partials + fused reduce/finalize (gw, gb, k1/k2 into ws)
 void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
                          const float* save_mean, const float* save_invstd,
-                         float* ws, float* part, int64_t rows, int C,
-                         bool relu, DT dt, hipStream_t s);
-void bn_bwd_finalize_launch(float* ws, const float* weight,
-                            const float* save_invstd, float* gw, float* gb,
-                            int64_t rows, int C, bool training, hipStream_t s);
+                         float* ws, float* part, float* gw, float* gb,
+                         int64_t rows, int C, bool relu, bool training, DT dt,
+                         hipStream_t s);
 void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
                          const float* save_mean, const float* save_invstd,
                          const float* weight, const float* ws, void* gx,
