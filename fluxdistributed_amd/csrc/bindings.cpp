@@ -149,6 +149,41 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
     return {gx, gw, gb};
 }
 
+std::tuple<at::Tensor, at::Tensor> maxpool_fwd(at::Tensor x, int64_t KH,
+                                               int64_t KW, int64_t S,
+                                               int64_t P) {
+    auto [rows, C] = nhwc_rows(x);
+    (void)rows;
+    const int V = dt_of(x) == DT::BF16 ? 8 : 4;
+    TORCH_CHECK(C % V == 0);
+    const int N = (int)x.size(0), H = (int)x.size(2), W = (int)x.size(3);
+    const int HO = (int)((H + 2 * P - KH) / S + 1);
+    const int WO = (int)((W + 2 * P - KW) / S + 1);
+    auto out = at::empty({N, C, HO, WO}, x.options())
+                   .contiguous(at::MemoryFormat::ChannelsLast);
+    auto idx = at::empty({(int64_t)N * HO * WO * C},
+                         x.options().dtype(at::kByte));
+    fda::maxpool_fwd_launch(x.data_ptr(), out.data_ptr(),
+                            idx.data_ptr<uint8_t>(), N, H, W, (int)C, HO, WO,
+                            (int)KH, (int)KW, (int)S, (int)P, dt_of(x),
+                            cur_stream());
+    return {out, idx};
+}
+
+at::Tensor maxpool_bwd(at::Tensor gout, at::Tensor idx, int64_t H, int64_t W,
+                       int64_t KH, int64_t KW, int64_t S, int64_t P) {
+    auto gc = gout.contiguous(at::MemoryFormat::ChannelsLast);
+    const int N = (int)gc.size(0), C = (int)gc.size(1);
+    const int HO = (int)gc.size(2), WO = (int)gc.size(3);
+    auto gx = at::empty({N, C, (int)H, (int)W}, gc.options())
+                  .contiguous(at::MemoryFormat::ChannelsLast);
+    fda::maxpool_bwd_launch(gc.data_ptr(), idx.data_ptr<uint8_t>(),
+                            gx.data_ptr(), N, (int)H, (int)W, C, HO, WO,
+                            (int)KH, (int)KW, (int)S, (int)P, dt_of(gc),
+                            cur_stream());
+    return gx;
+}
+
 void sgd_step(at::Tensor P, at::Tensor G, at::Tensor M, at::Tensor V,
               double lr, double mom, double wd, bool nesterov) {
     TORCH_CHECK(P.is_contiguous() && G.is_contiguous() && V.is_contiguous());
@@ -184,6 +219,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("add_relu_bwd", &add_relu_bwd);
     m.def("bn_act_fwd", &bn_act_fwd);
     m.def("bn_act_bwd", &bn_act_bwd);
+    m.def("maxpool_fwd", &maxpool_fwd);
+    m.def("maxpool_bwd", &maxpool_bwd);
     m.def("sgd_step", &sgd_step);
     m.def("adam_step", &adam_step);
     m.attr("_built_for") = "gfx950";

@@ -55,6 +55,14 @@ void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
                          int64_t rows, int C, bool relu, bool training, DT dt,
                          hipStream_t s);
 
+// MaxPool2d NHWC with saved argmax byte per output element
+void maxpool_fwd_launch(const void* x, void* out, uint8_t* idx, int N, int H,
+                        int W, int C, int HO, int WO, int KH, int KW, int S,
+                        int P, DT dt, hipStream_t s);
+void maxpool_bwd_launch(const void* gout, const uint8_t* idx, void* gx, int N,
+                        int H, int W, int C, int HO, int WO, int KH, int KW,
+                        int S, int P, DT dt, hipStream_t s);
+
 // flat fused optimizers. P: param dtype; M/V/S fp32; G param dtype.
 void sgd_step_launch(void* P, const void* G, float* M, float* V, int64_t n,
                      float lr, float mom, float wd, bool nesterov,

@@ -14,7 +14,7 @@ from typing import List, Optional, Type
 import torch
 import torch.nn as nn
 
-from ..ops.functional import batch_norm_act, fused_add_relu
+from ..ops.functional import batch_norm_act, fused_add_relu, MaxPool2d
 
 
 class FusedBNAct(nn.Module):
@@ -119,7 +119,7 @@ class ResNet(nn.Module):
             self.maxpool = nn.Identity()
         else:
             self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-            self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+            self.maxpool = MaxPool2d(3, stride=2, padding=1)
         self.bn1 = FusedBNAct(64, relu=True)
         self.layer1 = self._make_layer(block, 64, layers[0])
         self.layer2 = self._make_layer(block, 128, layers[1], stride=2)

@@ -14,5 +14,7 @@ from .functional import (  # noqa: F401
     logit_cross_entropy,
     fused_add_relu,
     batch_norm_act,
+    max_pool2d,
+    MaxPool2d,
 )
 from .fused_optim import FusedSGDMomentum, FusedAdam  # noqa: F401

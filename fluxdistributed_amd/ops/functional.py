@@ -91,6 +91,46 @@ def fused_add_relu(x: torch.Tensor, residual: torch.Tensor) -> torch.Tensor:
 
 
 # --------------------------------------------------------------------------
+# MaxPool2d NHWC with saved argmax (reference: ResNet stem 3x3 s2 pool;
+# SURVEY.md §2.4 "hand kernel + index mask").
+# --------------------------------------------------------------------------
+
+
+class _MaxPool2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel, stride, padding):
+        C = require_native("max_pool2d")
+        out, idx = C.maxpool_fwd(x, kernel, kernel, stride, padding)
+        ctx.save_for_backward(idx)
+        ctx.geom = (x.shape[2], x.shape[3], kernel, stride, padding)
+        return out
+
+    @staticmethod
+    def backward(ctx, gout):
+        (idx,) = ctx.saved_tensors
+        H, W, k, s, p = ctx.geom
+        C = require_native("max_pool2d")
+        gx = C.maxpool_bwd(gout, idx, H, W, k, k, s, p)
+        return gx, None, None, None
+
+
+def max_pool2d(x: torch.Tensor, kernel: int = 3, stride: int = 2,
+               padding: int = 1) -> torch.Tensor:
+    if _on_gpu(x):
+        return _MaxPool2d.apply(x, kernel, stride, padding)
+    return F.max_pool2d(x, kernel, stride, padding)
+
+
+class MaxPool2d(torch.nn.Module):
+    def __init__(self, kernel_size: int = 3, stride: int = 2, padding: int = 1):
+        super().__init__()
+        self.kernel_size, self.stride, self.padding = kernel_size, stride, padding
+
+    def forward(self, x):
+        return max_pool2d(x, self.kernel_size, self.stride, self.padding)
+
+
+# --------------------------------------------------------------------------
 # Fused BatchNorm (+ optional residual add) + ReLU, NHWC, training & eval.
 # Matches the reference semantic (SURVEY.md §7 hard-part 3): per-replica
 # running stats, never synced across data-parallel replicas.

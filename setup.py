@@ -21,6 +21,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "bindings.cpp"),
         os.path.join(CSRC, "elementwise.hip"),
         os.path.join(CSRC, "bn_act.hip"),
+        os.path.join(CSRC, "pool.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

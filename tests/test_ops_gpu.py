@@ -167,3 +167,29 @@ def test_fused_optimizer_end_to_end_gpu(seed):
         o2.step()
     for p1, p2 in zip(m1.parameters(), m2.parameters()):
         assert torch.allclose(p1, p2, rtol=1e-4, atol=1e-5)
+
+
+@pytest.mark.parametrize("dtype,tol", [(torch.float32, 0), (torch.bfloat16, 0)])
+@pytest.mark.parametrize("shape,k,s,p", [
+    ((2, 64, 112, 112), 3, 2, 1),   # ResNet stem pool
+    ((2, 128, 17, 17), 3, 2, 1),    # odd sizes
+    ((2, 64, 16, 16), 2, 2, 0),
+])
+def test_maxpool(dtype, tol, shape, k, s, p, seed):
+    C = _native()
+    x = torch.randn(*shape, device=DEV, dtype=dtype).contiguous(
+        memory_format=torch.channels_last)
+    out, idx = C.maxpool_fwd(x, k, k, s, p)
+    ref = F.max_pool2d(x.float(), k, s, p)
+    assert out.shape == ref.shape
+    assert torch.equal(out.float(), ref.to(out.dtype).float())
+    # backward: compare against autograd on the fp32 reference
+    xr = x.float().detach().requires_grad_()
+    refo = F.max_pool2d(xr, k, s, p)
+    g = torch.randn_like(refo)
+    refo.backward(g)
+    gx = C.maxpool_bwd(g.to(out.dtype), idx, shape[2], shape[3], k, k, s, p)
+    # ties may pick different argmax; tolerate tiny fraction of mismatches
+    diff = (gx.float() - xr.grad).abs()
+    frac_bad = float((diff > 1e-2).float().mean())
+    assert frac_bad < 2e-3, frac_bad
