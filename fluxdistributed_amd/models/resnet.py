@@ -37,10 +37,27 @@ class FusedBNAct(nn.Module):
         self.register_buffer("running_mean", torch.zeros(num_features))
         self.register_buffer("running_var", torch.ones(num_features))
         self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
+        # batch counter is kept host-side and flushed into the buffer lazily:
+        # a per-step GPU scalar add per BN layer was 29 launches/step of pure
+        # overhead (profiles/README.md)
+        self._nbt_pending = 0
+
+    def _flush_nbt(self):
+        if self._nbt_pending:
+            self.num_batches_tracked += self._nbt_pending
+            self._nbt_pending = 0
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        self._flush_nbt()
+        super()._save_to_state_dict(destination, prefix, keep_vars)
+
+    def _load_from_state_dict(self, *args, **kw):
+        self._nbt_pending = 0
+        super()._load_from_state_dict(*args, **kw)
 
     def forward(self, x: torch.Tensor, residual: Optional[torch.Tensor] = None):
         if self.training:
-            self.num_batches_tracked += 1
+            self._nbt_pending += 1
         return batch_norm_act(
             x, self.weight, self.bias, self.running_mean, self.running_var,
             self.training, self.momentum, self.eps, self.relu, residual,
