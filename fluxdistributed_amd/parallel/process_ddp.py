@@ -42,7 +42,10 @@ def init_process_group(backend: Optional[str] = None, timeout_s: int = 600):
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29531")
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = os.environ.get(
+            "FLUXDIST_PG_BACKEND",
+            "nccl" if torch.cuda.is_available() else "gloo",
+        )
     if backend == "nccl":
         torch.cuda.set_device(local_rank)
     dist.init_process_group(
