@@ -39,8 +39,11 @@ def parse_args():
     p.add_argument("--image-size", type=int, default=224)
     p.add_argument("--bucket-mb", type=float, default=25.0)
     p.add_argument("--no-overlap", action="store_true")
+    p.add_argument("--graph", action="store_true",
+                   help="capture the step in a hipGraph (measured slower than "
+                        "eager on ResNet-34 bs96 — see profiles/README.md)")
     p.add_argument("--no-graph", action="store_true",
-                   help="disable hipGraph step capture")
+                   help="(default) eager step; kept for compatibility")
     p.add_argument("--allow-cpu", action="store_true")
     p.add_argument("--lr", type=float, default=0.01)
     p.add_argument("--momentum", type=float, default=0.9)
@@ -101,10 +104,11 @@ def main():
 
     from fluxdistributed_amd.engine import make_train_step
 
-    example = next(loader) if (device.type == "cuda" and not args.no_graph) else None
+    use_graph = args.graph and not args.no_graph
+    example = next(loader) if (device.type == "cuda" and use_graph) else None
     train_step = make_train_step(model, opt, logit_cross_entropy,
                                  example_batch=example, ddp=ddp,
-                                 use_graph=not args.no_graph)
+                                 use_graph=use_graph)
 
     def step():
         x, y = next(loader)
