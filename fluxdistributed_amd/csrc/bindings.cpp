@@ -213,6 +213,57 @@ void adam_step(at::Tensor P, at::Tensor G, at::Tensor M, at::Tensor V,
 
 }  // namespace
 
+// ---- implicit-GEMM conv (NHWC bf16, conv_igemm.hip) -----------------------
+
+at::Tensor conv_igemm_fwd(at::Tensor x, at::Tensor w,
+                          int64_t sy, int64_t sx, int64_t py, int64_t px) {
+    TORCH_CHECK(x.dim() == 4 && w.dim() == 4);
+    TORCH_CHECK(x.scalar_type() == at::kBFloat16 && w.scalar_type() == at::kBFloat16,
+                "conv_igemm: bf16 only");
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "conv_igemm: x must be channels_last");
+    TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "conv_igemm: w must be channels_last");
+    const int N = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+              W = (int)x.size(3);
+    const int K = (int)w.size(0), R = (int)w.size(2), S = (int)w.size(3);
+    TORCH_CHECK((int)w.size(1) == C, "channel mismatch");
+    TORCH_CHECK(C % 64 == 0 && K % 64 == 0,
+                "conv_igemm: C and K must be multiples of 64");
+    const int P = (H + 2 * (int)py - R) / (int)sy + 1;
+    const int Q = (W + 2 * (int)px - S) / (int)sx + 1;
+    auto y = at::empty({N, K, P, Q},
+                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    fda::conv_igemm_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                           N, H, W, C, K, P, Q, R, S, (int)sy, (int)sx,
+                           (int)py, (int)px, /*dgrad=*/false, cur_stream());
+    return y;
+}
+
+at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
+                            int64_t C, int64_t H, int64_t W,
+                            int64_t R, int64_t S,
+                            int64_t sy, int64_t sx, int64_t py, int64_t px) {
+    // dy: [N,K,P,Q] channels_last; wt: [R*S*C, K] row-major (pre-transposed
+    // weight, k contiguous). Output dx: [N,C,H,W] channels_last.
+    TORCH_CHECK(dy.dim() == 4 && dy.scalar_type() == at::kBFloat16);
+    TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast));
+    TORCH_CHECK(wt.dim() == 2 && wt.is_contiguous() &&
+                wt.scalar_type() == at::kBFloat16);
+    const int N = (int)dy.size(0), K = (int)dy.size(1), P = (int)dy.size(2),
+              Q = (int)dy.size(3);
+    TORCH_CHECK(wt.size(0) == R * S * C && wt.size(1) == K, "wt shape mismatch");
+    TORCH_CHECK(C % 64 == 0 && K % 64 == 0,
+                "conv_igemm: C and K must be multiples of 64");
+    auto dx = at::empty({N, C, H, W},
+                        dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+    fda::conv_igemm_launch(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
+                           N, (int)H, (int)W, (int)C, K, P, Q, (int)R, (int)S,
+                           (int)sy, (int)sx, (int)py, (int)px, /*dgrad=*/true,
+                           cur_stream());
+    return dx;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ce_fwd", &ce_fwd, "fused logit cross-entropy fwd (loss + dlogits)");
     m.def("add_relu_fwd", &add_relu_fwd);
@@ -223,5 +274,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("maxpool_bwd", &maxpool_bwd);
     m.def("sgd_step", &sgd_step);
     m.def("adam_step", &adam_step);
+    m.def("conv_igemm_fwd", &conv_igemm_fwd,
+          "implicit-GEMM conv fwd (NHWC bf16, MFMA)");
+    m.def("conv_igemm_dgrad", &conv_igemm_dgrad,
+          "implicit-GEMM conv input-grad (NHWC bf16, MFMA)");
     m.attr("_built_for") = "gfx950";
 }

@@ -63,6 +63,14 @@ void maxpool_bwd_launch(const void* gout, const uint8_t* idx, void* gx, int N,
                         int H, int W, int C, int HO, int WO, int KH, int KW,
                         int S, int P, DT dt, hipStream_t s);
 
+// Implicit-GEMM conv, NHWC bf16 only (see conv_igemm.hip).
+// fwd:   src=x [N,H,W,C], wgt=w [K][R*S*C], out=y [N*P*Q][K]
+// dgrad: src=dy [N,P,Q,K], wgt=wt [R*S*C][K] (pre-transposed), out=dx
+void conv_igemm_launch(const void* src, const void* wgt, void* out,
+                       int N, int H, int W, int C, int K, int P, int Q,
+                       int R, int S, int sy, int sx, int py, int px,
+                       bool dgrad, hipStream_t stream);
+
 // flat fused optimizers. P: param dtype; M/V/S fp32; G param dtype.
 void sgd_step_launch(void* P, const void* G, float* M, float* V, int64_t n,
                      float lr, float mom, float wd, bool nesterov,

@@ -65,11 +65,15 @@ class FusedBNAct(nn.Module):
 
 
 def conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
-    return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
+    from ..ops.conv import FdaConv2d
+
+    return FdaConv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
 
 
 def conv1x1(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
-    return nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
+    from ..ops.conv import FdaConv2d
+
+    return FdaConv2d(cin, cout, 1, stride=stride, bias=False)
 
 
 class BasicBlock(nn.Module):

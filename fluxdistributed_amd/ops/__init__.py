@@ -18,3 +18,4 @@ from .functional import (  # noqa: F401
     MaxPool2d,
 )
 from .fused_optim import FusedSGDMomentum, FusedAdam  # noqa: F401
+from .conv import fda_conv2d, FdaConv2d  # noqa: F401

@@ -1,0 +1,100 @@
+"""Conv2d dispatch: hand-written MFMA implicit-GEMM kernels with library
+fallback.
+
+The reference gets conv from cuDNN via NNlibCUDA; the
+MI355X-native path is fluxdistributed_amd/csrc/conv_igemm.hip —
+v_mfma_f32_16x16x32_bf16 tiles, LDS-staged via global_load_lds. Supported
+there: NHWC bf16, dilation 1, groups 1, C and K multiples of 64 (all
+ResNet body convs). The stem (C=3) and anything else goes to the library
+path (MIOpen through torch), and wgrad currently does too.
+
+Env:
+  FLUXDIST_CONV=miopen   force the library path everywhere (A/B testing)
+  FLUXDIST_CONV=fda      force the native path (errors on unsupported)
+"""
+
+import os
+from typing import Tuple
+
+import torch
+import torch.nn.functional as F
+
+from .native import load_native, require_native
+
+
+def _pair(v) -> Tuple[int, int]:
+    return (v, v) if isinstance(v, int) else tuple(v)
+
+
+def _native_supported(x: torch.Tensor, w: torch.Tensor, stride, padding,
+                      dilation, groups) -> bool:
+    if os.environ.get("FLUXDIST_CONV", "") == "miopen":
+        return False
+    if not (x.is_cuda and x.dtype == torch.bfloat16 and w.dtype == torch.bfloat16):
+        return False
+    if groups != 1 or _pair(dilation) != (1, 1):
+        return False
+    C, K = x.shape[1], w.shape[0]
+    if C % 64 != 0 or K % 64 != 0:
+        return False
+    return load_native() is not None
+
+
+class _FdaConv2d(torch.autograd.Function):
+    """Forward + input-grad on the native implicit-GEMM kernels; weight-grad
+    via the library (aten convolution_backward with weight-only mask)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, padding):
+        C = require_native("conv_igemm_fwd")
+        xc = x.contiguous(memory_format=torch.channels_last)
+        wc = weight.contiguous(memory_format=torch.channels_last)
+        sy, sx = stride
+        py, px = padding
+        y = C.conv_igemm_fwd(xc, wc, sy, sx, py, px)
+        ctx.save_for_backward(xc, wc)
+        ctx.conf = (stride, padding)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, w = ctx.saved_tensors
+        stride, padding = ctx.conf
+        sy, sx = stride
+        py, px = padding
+        gy = gy.contiguous(memory_format=torch.channels_last)
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            C = require_native("conv_igemm_dgrad")
+            K, Cin, R, S = w.shape
+            # wt[(r*S+s)*C + c][k]: k-contiguous rows for the staged B tile
+            wt = w.permute(2, 3, 1, 0).reshape(R * S * Cin, K).contiguous()
+            dx = C.conv_igemm_dgrad(gy, wt, Cin, x.shape[2], x.shape[3],
+                                    R, S, sy, sx, py, px)
+        if ctx.needs_input_grad[1]:
+            dw = torch.ops.aten.convolution_backward(
+                gy, x, w, None, list(stride), list(padding), [1, 1],
+                False, [0, 0], 1, [False, True, False])[1]
+        return dx, dw, None, None
+
+
+def fda_conv2d(x: torch.Tensor, weight: torch.Tensor, stride=(1, 1),
+               padding=(0, 0), dilation=(1, 1), groups: int = 1) -> torch.Tensor:
+    """conv2d with per-shape dispatch to the native MFMA kernel."""
+    stride, padding, dilation = _pair(stride), _pair(padding), _pair(dilation)
+    if _native_supported(x, weight, stride, padding, dilation, groups):
+        return _FdaConv2d.apply(x, weight, stride, padding)
+    if os.environ.get("FLUXDIST_CONV", "") == "fda" and x.is_cuda:
+        raise RuntimeError(
+            f"FLUXDIST_CONV=fda but shape unsupported by conv_igemm: "
+            f"x={tuple(x.shape)} w={tuple(weight.shape)} groups={groups}")
+    return F.conv2d(x, weight, None, stride, padding, dilation, groups)
+
+
+class FdaConv2d(torch.nn.Conv2d):
+    """nn.Conv2d whose forward goes through fda_conv2d (bias-free, as the
+    reference's Flux convs are — /root/reference README.md model usage)."""
+
+    def forward(self, x):
+        return fda_conv2d(x, self.weight, self.stride, self.padding,
+                          self.dilation, self.groups)

@@ -22,6 +22,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "elementwise.hip"),
         os.path.join(CSRC, "bn_act.hip"),
         os.path.join(CSRC, "pool.hip"),
+        os.path.join(CSRC, "conv_igemm.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
