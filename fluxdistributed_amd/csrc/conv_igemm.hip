@@ -4,9 +4,8 @@
 // from cuDNN via NNlibCUDA (/root/reference -> Flux conv):
 // hand-written MFMA kernels — v_mfma_f32_16x16x32_bf16 tiles, LDS staging
 // via global_load_lds (direct HBM->LDS DMA), source-side XOR swizzle for
-// bank-conflict-free ds_read_b128 fragment reads (guide T2 / rule 21), and
-// a 3-buffer counted-vmcnt pipeline (guide T3/T4: glds for K-step t+2 stays
-// in flight across the barrier while K-step t computes).
+// bank-conflict-free ds_read_b128 fragment reads (guide T2 / rule 21),
+// double-buffered with the staging DMA in flight across the compute phase.
 //
 // GEMM view (forward):
 //   M = N*P*Q output pixels, Nd = K output channels, Kd = R*S*C
@@ -23,13 +22,16 @@
 // for 3x3 stride-2). B = pre-transposed weights wt[rs*C + c][k]
 // (k-contiguous rows).
 //
-// Tiles: BM=128 x BN=64 x BK=64, 256 threads (4 waves as 2x2), per-wave
-// 64x32 output = 4x2 fragments of 16x16, fp32 accumulate, bf16 store.
-// LDS: (128*64 + 64*64) bf16 * 3 buffers = 72 KiB -> 2 blocks/CU.
+// Tile geometry (templated; wave tile fixed at 64x64 = 4x4 fragments of
+// 16x16 so every config runs 32 MFMAs per K-step per wave — the
+// MFMA-per-glds ratio is what sets throughput, guide §5 ladder):
+//   OC % 128 == 0 : BM=128 x BN=128, 4 waves as 2x2, LDS 2x32 KiB
+//   OC % 128 != 0 : BM=256 x BN=64,  4 waves as 4x1, LDS 2x40 KiB
+// fp32 accumulate, bf16 store.
 //
 // Constraints (host wrapper): staged reduction channels (C fwd / K dgrad)
-// and output channels both multiples of 64, dilation 1, groups 1. The
-// ResNet stem (C=3) falls back to the library path.
+// and output channels multiples of 64, dilation 1, groups 1. The ResNet
+// stem (C=3) falls back to the library path.
 
 #include <hip/hip_runtime.h>
 #include "fda_common.h"
@@ -48,42 +50,35 @@ __device__ __align__(16) static const unsigned short conv_zero16[8] = {0};
 
 enum ConvMode { CONV_FWD = 0, CONV_DGRAD = 1 };
 
-constexpr int BM = 128, BN = 64, BK = 64;
-constexpr int A_ELEMS = BM * BK;          // 8192 bf16 = 16 KiB
-constexpr int B_ELEMS = BN * BK;          // 4096 bf16 = 8 KiB
-constexpr int BUF_ELEMS = A_ELEMS + B_ELEMS;
-constexpr int NBUF = 3;                   // glds tile ring
-// glds instructions issued per K-step tile pair (A: 16, B: 8, split over
-// 4 waves). vmcnt is per-wave: 6 per wave per tile.
-constexpr int GLDS_PER_WAVE = 6;
+constexpr int BK = 64;
 
-template <int MODE>
+template <int MODE, int BM, int BN, int WN>
 __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
-    const unsigned short* __restrict__ src,   // x (fwd) / dy (dgrad), NHWC
-    const unsigned short* __restrict__ wgt,   // w [K][RS*C] (fwd) / wt [RS*C][K] (dgrad)
-    unsigned short* __restrict__ out,         // y [M][K] (fwd) / dx NHWC (dgrad)
-    int N, int H, int W, int C,               // input tensor dims (fwd view)
-    int K, int P, int Q,                      // output channels & spatial
+    const unsigned short* __restrict__ src,
+    const unsigned short* __restrict__ wgt,
+    unsigned short* __restrict__ out,
+    int N, int H, int W, int C,
+    int K, int P, int Q,
     int R, int S, int sy, int sx, int py, int px) {
-    const int OC = (MODE == CONV_FWD) ? K : C;   // Nd of the GEMM
-    const int RC = (MODE == CONV_FWD) ? C : K;   // staged reduction channels
+    constexpr int A_ELEMS = BM * BK;
+    constexpr int B_ELEMS = BN * BK;
+    constexpr int BUF_ELEMS = A_ELEMS + B_ELEMS;
+    constexpr int AI = BM / 32;          // A glds per wave per tile
+    constexpr int BI = BN / 32;          // B glds per wave per tile
 
-    // ---- per-class output-pixel space ------------------------------------
-    // fwd: one class, pixels (n,p,q). dgrad: class (a,b) = (h%sy, w%sx),
-    // pixels (n, h'=h/sy, w'=w/sx) with h = a + sy*h'.
+    const int OC = (MODE == CONV_FWD) ? K : C;
+    const int RC = (MODE == CONV_FWD) ? C : K;
+
     int a = 0, b = 0, OH, OW, r0 = 0, s0 = 0, nR = R, nS = S;
     if (MODE == CONV_FWD) {
         OH = P; OW = Q;
     } else {
         a = blockIdx.z / sx;  b = blockIdx.z % sx;
-        OH = (H - a + sy - 1) / sy;          // # h' values
+        OH = (H - a + sy - 1) / sy;
         OW = (W - b + sx - 1) / sx;
-        // valid taps: r == (a+py) mod sy, s == (b+px) mod sx
         r0 = (a + py) % sy;  nR = (R - r0 + sy - 1) / sy;
         s0 = (b + px) % sx;  nS = (S - s0 + sx - 1) / sx;
         if (OH <= 0 || OW <= 0) return;
-        // a class with no valid taps still writes its (all-zero) pixels:
-        // nR/nS <= 0 makes T = 0 below and the epilogue stores zero acc.
         if (nR < 0) nR = 0;
         if (nS < 0) nS = 0;
     }
@@ -95,20 +90,20 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wid = tid >> 6;
-    const int wm = wid >> 1;
-    const int wn = wid & 1;
+    const int wm = wid / WN;             // wave row (64-pixel granularity)
+    const int wn = wid % WN;             // wave col (64-channel granularity)
 
-    extern __shared__ unsigned short lds[];   // [NBUF][BUF_ELEMS]
+    extern __shared__ unsigned short lds[];   // [2][BUF_ELEMS]
 
     // ---- per-lane staging descriptors ------------------------------------
-    int a_row[4];
-    long a_pix[4];
-    int a_hb[4], a_wb[4];
-    bool a_mok[4];
+    int a_row[AI];
+    long a_pix[AI];
+    int a_hb[AI], a_wb[AI];
+    bool a_mok[AI];
     const int cslot = lane & 7;
     #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-        const int row = (wid * 4 + i) * 8 + (lane >> 3);
+    for (int i = 0; i < AI; ++i) {
+        const int row = (wid * AI + i) * 8 + (lane >> 3);
         a_row[i] = row;
         const long m = m0 + row;
         const bool mok = m < M;
@@ -122,16 +117,14 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
             a_wb[i] = ow * sx - px;
             a_pix[i] = ((long)n * H) * W * C;
         } else {
-            // h = a + sy*oh; p = (h + py - r)/sy = oh + (a+py-r)/sy (exact
-            // for the class's taps). Precompute p,q bases.
-            a_hb[i] = oh + (a + py) / sy;   // p for r = r0 (subtract ri later)
+            a_hb[i] = oh + (a + py) / sy;
             a_wb[i] = ow + (b + px) / sx;
             a_pix[i] = ((long)n * P) * Q * K;
         }
     }
-    int b_row[2];
+    int b_row[BI];
     #pragma unroll
-    for (int i = 0; i < 2; ++i) b_row[i] = (wid * 2 + i) * 8 + (lane >> 3);
+    for (int i = 0; i < BI; ++i) b_row[i] = (wid * BI + i) * 8 + (lane >> 3);
 
     const int cblocks = RC / BK;
     const int T = nR * nS * cblocks;
@@ -142,14 +135,14 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
         const int ri = rsi / nS, si = rsi % nS;
         unsigned short* base = lds + buf * BUF_ELEMS;
         #pragma unroll
-        for (int i = 0; i < 4; ++i) {
+        for (int i = 0; i < AI; ++i) {
             const int row = a_row[i];
             const int cs = (cslot ^ (row & 7)) * 8;
             const unsigned short* sp;
             bool ok = a_mok[i];
             long off = 0;
             if (MODE == CONV_FWD) {
-                const int h = a_hb[i] + ri, w = a_wb[i] + si;   // nR=R,nS=S
+                const int h = a_hb[i] + ri, w = a_wb[i] + si;
                 ok = ok && (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W;
                 off = a_pix[i] + ((long)h * W + w) * C + cb + cs;
             } else {
@@ -158,13 +151,13 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
                 off = a_pix[i] + ((long)p * Q + q) * K + cb + cs;
             }
             sp = ok ? src + off : conv_zero16;
-            FDA_GLDS16(sp, base + (wid * 4 + i) * 8 * BK);
+            FDA_GLDS16(sp, base + (wid * AI + i) * 8 * BK);
         }
         const int r = r0 + ri * ((MODE == CONV_FWD) ? 1 : sy);
         const int s = s0 + si * ((MODE == CONV_FWD) ? 1 : sx);
         const int rs = r * S + s;
         #pragma unroll
-        for (int i = 0; i < 2; ++i) {
+        for (int i = 0; i < BI; ++i) {
             const int row = b_row[i];
             const int cs = (cslot ^ (row & 7)) * 8;
             const unsigned short* sp;
@@ -173,12 +166,12 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
             } else {
                 sp = wgt + ((long)((long)rs * C + n0 + row) * K + cb + cs);
             }
-            FDA_GLDS16(sp, base + A_ELEMS + (wid * 2 + i) * 8 * BK);
+            FDA_GLDS16(sp, base + A_ELEMS + (wid * BI + i) * 8 * BK);
         }
     };
 
     // ---- fragment read offsets (elements into an lds buffer) -------------
-    int a_off[4][2], b_off[2][2];
+    int a_off[4][2], b_off[4][2];
     {
         const int fr = lane & 15, fq = lane >> 4;
         #pragma unroll
@@ -190,43 +183,37 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
                 a_off[mi][kh] = row * BK + slot * 8;
             }
         #pragma unroll
-        for (int ni = 0; ni < 2; ++ni)
+        for (int ni = 0; ni < 4; ++ni)
             #pragma unroll
             for (int kh = 0; kh < 2; ++kh) {
-                const int row = wn * 32 + ni * 16 + fr;
+                const int row = wn * 64 + ni * 16 + fr;
                 const int slot = (kh * 4 + fq) ^ (row & 7);
                 b_off[ni][kh] = A_ELEMS + row * BK + slot * 8;
             }
     }
 
-    floatx4 acc[4][2];
+    floatx4 acc[4][4];
     #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
         #pragma unroll
-        for (int ni = 0; ni < 2; ++ni) acc[mi][ni] = floatx4{0.f, 0.f, 0.f, 0.f};
+        for (int ni = 0; ni < 4; ++ni) acc[mi][ni] = floatx4{0.f, 0.f, 0.f, 0.f};
 
-    // ---- main loop: 3-buffer ring, counted vmcnt (guide T3+T4) -----------
-    // stage t and t+1 up front; inside the loop tile t+2's DMA stays in
-    // flight across the barrier (vmcnt(GLDS_PER_WAVE) = "previous tile's
-    // loads may still be outstanding, mine have landed").
+    // ---- main loop: double buffer; tile t+1's DMA in flight over tile t's
+    // compute, drained at the iteration boundary (guide T3 minimum form) ---
     if (T > 0) stage(0, 0);
-    if (T > 1) stage(1, 1);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
     for (int it = 0; it < T; ++it) {
-        if (it + 1 < T)
-            asm volatile("s_waitcnt vmcnt(6)" ::: "memory");  // GLDS_PER_WAVE
-        else
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
-        if (it + 2 < T) stage((it + 2) % NBUF, it + 2);
-        const unsigned short* buf = lds + (it % NBUF) * BUF_ELEMS;
-        short8 af[4][2], bf[2][2];
+        if (it + 1 < T) stage((it + 1) & 1, it + 1);
+        const unsigned short* buf = lds + (it & 1) * BUF_ELEMS;
+        short8 af[4][2], bf[4][2];
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             #pragma unroll
             for (int kh = 0; kh < 2; ++kh)
                 af[mi][kh] = *(const short8*)(buf + a_off[mi][kh]);
         #pragma unroll
-        for (int ni = 0; ni < 2; ++ni)
+        for (int ni = 0; ni < 4; ++ni)
             #pragma unroll
             for (int kh = 0; kh < 2; ++kh)
                 bf[ni][kh] = *(const short8*)(buf + b_off[ni][kh]);
@@ -236,13 +223,11 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
             #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
                 #pragma unroll
-                for (int ni = 0; ni < 2; ++ni)
+                for (int ni = 0; ni < 4; ++ni)
                     acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         af[mi][kh], bf[ni][kh], acc[mi][ni], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
-        // reads of buf[it] complete before the next iteration's barrier
-        // lets anyone overwrite it (ds_read results consumed by the MFMAs;
-        // lgkm waits are compiler-inserted before each use).
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
     }
 
@@ -258,19 +243,36 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
             if (MODE == CONV_FWD) {
                 obase = m * OC;
             } else {
-                // m -> (n, h', w') -> (n, a + sy*h', b + sx*w')
                 const int ww = (int)(m % OW);
                 const int hh = (int)((m / OW) % OH);
                 const int n = (int)(m / ((long)OW * OH));
                 obase = (((long)n * H + a + (long)sy * hh) * W + b +
                          (long)sx * ww) * C;
             }
-            unsigned short* orow = out + obase + n0 + wn * 32;
+            unsigned short* orow = out + obase + n0 + wn * 64;
             #pragma unroll
-            for (int ni = 0; ni < 2; ++ni)
+            for (int ni = 0; ni < 4; ++ni)
                 orow[ni * 16 + fcol] = f32_to_bf16bits(acc[mi][ni][j]);
         }
     }
+}
+
+template <int MODE, int BM, int BN, int WN>
+static void launch_cfg(const void* src, const void* wgt, void* out,
+                       int N, int H, int W, int C, int K, int P, int Q,
+                       int R, int S, int sy, int sx, int py, int px,
+                       hipStream_t stream) {
+    const int OC = (MODE == CONV_FWD) ? K : C;
+    const long M = (MODE == CONV_FWD)
+        ? (long)N * P * Q
+        : (long)N * ((H + sy - 1) / sy) * ((W + sx - 1) / sx);
+    dim3 grid((unsigned)((M + BM - 1) / BM), (unsigned)(OC / BN),
+              (MODE == CONV_FWD) ? 1u : (unsigned)(sy * sx));
+    const size_t shmem = 2 * (BM * BK + BN * BK) * sizeof(unsigned short);
+    hipLaunchKernelGGL((conv_igemm_kernel<MODE, BM, BN, WN>), grid, dim3(256),
+                       shmem, stream, (const unsigned short*)src,
+                       (const unsigned short*)wgt, (unsigned short*)out,
+                       N, H, W, C, K, P, Q, R, S, sy, sx, py, px);
 }
 
 void conv_igemm_launch(const void* src, const void* wgt, void* out,
@@ -278,25 +280,24 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int R, int S, int sy, int sx, int py, int px,
                        bool dgrad, hipStream_t stream) {
     const int OC = dgrad ? C : K;
-    dim3 block(256);
-    const size_t shmem = NBUF * BUF_ELEMS * sizeof(unsigned short);
     if (dgrad) {
-        // per-parity-class pixel count varies; grid.x sized for the largest
-        // class (a=b=0: ceil(H/sy)*ceil(W/sx)); smaller classes early-return.
-        const long Mcls = (long)N * ((H + sy - 1) / sy) * ((W + sx - 1) / sx);
-        dim3 grid((unsigned)((Mcls + BM - 1) / BM), (unsigned)(OC / BN),
-                  (unsigned)(sy * sx));
-        hipLaunchKernelGGL((conv_igemm_kernel<CONV_DGRAD>), grid, block, shmem,
-                           stream, (const unsigned short*)src,
-                           (const unsigned short*)wgt, (unsigned short*)out,
-                           N, H, W, C, K, P, Q, R, S, sy, sx, py, px);
+        if (OC % 128 == 0)
+            launch_cfg<CONV_DGRAD, 128, 128, 2>(src, wgt, out, N, H, W, C, K,
+                                                P, Q, R, S, sy, sx, py, px,
+                                                stream);
+        else
+            launch_cfg<CONV_DGRAD, 256, 64, 1>(src, wgt, out, N, H, W, C, K,
+                                               P, Q, R, S, sy, sx, py, px,
+                                               stream);
     } else {
-        const long M = (long)N * P * Q;
-        dim3 grid((unsigned)((M + BM - 1) / BM), (unsigned)(OC / BN));
-        hipLaunchKernelGGL((conv_igemm_kernel<CONV_FWD>), grid, block, shmem,
-                           stream, (const unsigned short*)src,
-                           (const unsigned short*)wgt, (unsigned short*)out,
-                           N, H, W, C, K, P, Q, R, S, sy, sx, py, px);
+        if (OC % 128 == 0)
+            launch_cfg<CONV_FWD, 128, 128, 2>(src, wgt, out, N, H, W, C, K,
+                                              P, Q, R, S, sy, sx, py, px,
+                                              stream);
+        else
+            launch_cfg<CONV_FWD, 256, 64, 1>(src, wgt, out, N, H, W, C, K,
+                                             P, Q, R, S, sy, sx, py, px,
+                                             stream);
     }
 }
 
