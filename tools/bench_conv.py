@@ -48,6 +48,10 @@ def main():
     p.add_argument("--batch", type=int, default=96)
     p.add_argument("--iters", type=int, default=50)
     p.add_argument("--mode", choices=["fwd", "dgrad", "both"], default="both")
+    p.add_argument("--only", type=int, default=-1,
+                   help="run only SHAPES[i] (for PMC profiling runs)")
+    p.add_argument("--fda-only", action="store_true",
+                   help="skip the library arms (clean kernel-trace)")
     args = p.parse_args()
     assert torch.cuda.is_available()
     torch.backends.cudnn.benchmark = True
@@ -58,7 +62,8 @@ def main():
     print(f"{'shape':34s} {'fda_fwd':>9s} {'lib_fwd':>9s} {'fda_TF':>7s} {'lib_TF':>7s}"
           f" | {'fda_dg':>9s} {'lib_dg':>9s}")
     tot_fda_f = tot_lib_f = tot_fda_d = tot_lib_d = 0.0
-    for (c, h, w, k, r, s, cnt) in SHAPES:
+    shapes = SHAPES if args.only < 0 else [SHAPES[args.only]]
+    for (c, h, w, k, r, s, cnt) in shapes:
         pad = r // 2
         P, Q = (h + 2 * pad - r) // s + 1, (w + 2 * pad - r) // s + 1
         x = torch.randn(n, c, h, w, device="cuda").bfloat16() \
@@ -68,8 +73,8 @@ def main():
         flops = 2.0 * n * P * Q * k * c * r * r
 
         t_ff = timeit(lambda: C_.conv_igemm_fwd(x, wt, s, s, pad, pad), args.iters)
-        t_lf = timeit(lambda: torch.nn.functional.conv2d(x, wt, None, s, pad),
-                      args.iters)
+        t_lf = (timeit(lambda: torch.nn.functional.conv2d(x, wt, None, s, pad),
+                       args.iters) if not args.fda_only else float("inf"))
         gy = torch.randn(n, k, P, Q, device="cuda").bfloat16() \
             .contiguous(memory_format=torch.channels_last)
         wtt = wt.permute(2, 3, 1, 0).reshape(r * r * c, k).contiguous()
@@ -81,7 +86,7 @@ def main():
             return torch.ops.aten.convolution_backward(
                 gy, xg, wt, None, [s, s], [pad, pad], [1, 1], False, [0, 0],
                 1, [True, False, False])[0]
-        t_ld = timeit(lib_dgrad, args.iters)
+        t_ld = timeit(lib_dgrad, args.iters) if not args.fda_only else float("inf")
 
         name = f"{c}x{h}x{w} k{k} {r}x{r} s{s} x{cnt}"
         print(f"{name:34s} {t_ff*1e6:9.1f} {t_lf*1e6:9.1f} "

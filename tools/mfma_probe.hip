@@ -36,6 +36,20 @@ __global__ void probe(const float* A, const float* B, float* C){
     for (int j=0;j<4;++j) C[(r0+j)*16 + col] = acc[j];
 }
 
+// ---- probe 2: ds_read_tr16_b64 gather map ---------------------------------
+// LDS filled with lds[e] = e; each lane reads at addr = base + lane*8B and
+// we print which element index each (lane, j) slot received.
+typedef __attribute__((ext_vector_type(4))) short short4_;
+__global__ void probe_tr(unsigned short* out){
+    __shared__ unsigned short lds[512];
+    int l = threadIdx.x;
+    for (int i = l; i < 512; i += 64) lds[i] = (unsigned short)i;
+    __syncthreads();
+    short4_ v = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+        (__attribute__((address_space(3))) short4_*)&lds[l*4]);
+    for (int j=0;j<4;++j) out[l*4+j] = (unsigned short)v[j];
+}
+
 int main(){
     float *A, *B, *C;
     hipMallocManaged(&A, 16*32*4); hipMallocManaged(&B, 32*16*4); hipMallocManaged(&C, 16*16*4);
@@ -46,6 +60,11 @@ int main(){
     auto bf=[&](float f){unsigned u;memcpy(&u,&f,4);unsigned l=(u>>16)&1u;u+=0x7fff+l;u=(u>>16)<<16;float r;memcpy(&r,&u,4);return r;};
     float ref[256];
     for(int i=0;i<16;++i)for(int j=0;j<16;++j){float s=0;for(int k=0;k<32;++k)s+=bf(A[i*32+k])*bf(B[k*16+j]);ref[i*16+j]=s;}
+    unsigned short* T; hipMallocManaged(&T, 64*4*2);
+    hipLaunchKernelGGL(probe_tr, dim3(1), dim3(64), 0, 0, T);
+    hipDeviceSynchronize();
+    printf("tr16 map (lane: e0 e1 e2 e3):\n");
+    for(int l=0;l<64;++l){printf("%2d: %3d %3d %3d %3d%s",l,T[l*4],T[l*4+1],T[l*4+2],T[l*4+3], (l%4==3)?"\n":"   ");}
     for (int MAP=0; MAP<2; ++MAP){
         if (MAP==0) hipLaunchKernelGGL(probe<0>, dim3(1), dim3(64), 0, 0, A,B,C);
         else        hipLaunchKernelGGL(probe<1>, dim3(1), dim3(64), 0, 0, A,B,C);
@@ -55,3 +74,4 @@ int main(){
     }
     return 0;
 }
+
