@@ -264,6 +264,39 @@ at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
     return dx;
 }
 
+at::Tensor conv_igemm_wgrad(at::Tensor dy, at::Tensor x,
+                            int64_t R, int64_t S,
+                            int64_t sy, int64_t sx, int64_t py, int64_t px) {
+    // dy: [N,K,P,Q] channels_last bf16; x: [N,C,H,W] channels_last bf16.
+    // Returns ws[K][R*S*C] fp32 (the channels_last weight-grad memory
+    // layout [K][R][S][C] flattened).
+    TORCH_CHECK(dy.scalar_type() == at::kBFloat16 &&
+                x.scalar_type() == at::kBFloat16);
+    TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                x.is_contiguous(at::MemoryFormat::ChannelsLast));
+    const int N = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+              W = (int)x.size(3);
+    const int K = (int)dy.size(1), P = (int)dy.size(2), Q = (int)dy.size(3);
+    TORCH_CHECK(C % 64 == 0 && K % 64 == 0);
+    auto ws = at::zeros({K, R * S * C}, x.options().dtype(at::kFloat));
+    fda::conv_wgrad_launch(dy.data_ptr(), x.data_ptr(), ws.data_ptr<float>(),
+                           N, H, W, C, K, P, Q, (int)R, (int)S, (int)sy,
+                           (int)sx, (int)py, (int)px, cur_stream());
+    return ws;
+}
+
+void wt_transpose_batch(at::Tensor src_ptrs, at::Tensor dst_ptrs,
+                        at::Tensor Ks, at::Tensor RCs, at::Tensor tile_counts,
+                        int64_t max_tiles) {
+    TORCH_CHECK(src_ptrs.is_cuda() && src_ptrs.scalar_type() == at::kLong);
+    const int n = (int)src_ptrs.numel();
+    fda::wt_transpose_launch(src_ptrs.data_ptr<int64_t>(),
+                             dst_ptrs.data_ptr<int64_t>(),
+                             Ks.data_ptr<int>(), RCs.data_ptr<int>(),
+                             tile_counts.data_ptr<int>(), n, (int)max_tiles,
+                             cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ce_fwd", &ce_fwd, "fused logit cross-entropy fwd (loss + dlogits)");
     m.def("add_relu_fwd", &add_relu_fwd);
@@ -276,6 +309,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("adam_step", &adam_step);
     m.def("conv_igemm_fwd", &conv_igemm_fwd,
           "implicit-GEMM conv fwd (NHWC bf16, MFMA)");
+    m.def("wt_transpose_batch", &wt_transpose_batch);
+    m.def("conv_igemm_wgrad", &conv_igemm_wgrad,
+          "implicit-GEMM conv weight-grad (NHWC bf16, MFMA + tr16 reads)");
     m.def("conv_igemm_dgrad", &conv_igemm_dgrad,
           "implicit-GEMM conv input-grad (NHWC bf16, MFMA)");
     m.attr("_built_for") = "gfx950";

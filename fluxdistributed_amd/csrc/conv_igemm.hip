@@ -302,3 +302,236 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
 }
 
 }  // namespace fda
+
+namespace fda {
+
+// ---- batched conv-weight transpose ----------------------------------------
+// One launch transposes every conv weight w[k][rs*C+c] -> wt[rs*C+c][k] for
+// the dgrad B-tiles (replaces 36 per-layer permute kernels per step).
+// Per tensor: 2D 64x64 LDS-tiled transpose, read coalesced along rc, write
+// coalesced along k. All conv weights have K and RS*C multiples of 64.
+__global__ __launch_bounds__(256) void wt_transpose_kernel(
+    const long* __restrict__ src_ptrs,   // device addresses of w tensors
+    long* __restrict__ dst_ptrs,         // device addresses of wt slices
+    const int* __restrict__ Ks, const int* __restrict__ RCs,
+    const int* __restrict__ tile_counts) {
+    const int t = blockIdx.y;
+    const int K = Ks[t], RC = RCs[t];
+    const int kt = K / 64, rt = RC / 64;
+    if ((int)blockIdx.x >= tile_counts[t]) return;
+    const int tk = blockIdx.x % kt, trc = blockIdx.x / kt;
+    (void)rt;
+    const unsigned short* src = (const unsigned short*)src_ptrs[t];
+    unsigned short* dst = (unsigned short*)dst_ptrs[t];
+
+    __shared__ unsigned short tile[64][64 + 8];  // +8 bf16 pad: no bank dup
+    const int tid = threadIdx.x;
+    // read: 64 k-rows x 64 rc; thread reads 16 elems, 8-contig along rc
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {
+        const int kk = (tid / 8) + i * 32;           // 0..63
+        const int rc = (tid % 8) * 8;                // 0..56 step 8
+        const unsigned short* sp =
+            src + (long)(tk * 64 + kk) * RC + trc * 64 + rc;
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) tile[kk][rc + e] = sp[e];
+    }
+    __syncthreads();
+    // write: 64 rc-rows x 64 k; thread writes 16 elems, 8-contig along k
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {
+        const int rc = (tid / 8) + i * 32;
+        const int kk = (tid % 8) * 8;
+        unsigned short* dp =
+            dst + (long)(trc * 64 + rc) * K + tk * 64 + kk;
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) dp[e] = tile[kk + e][rc];
+    }
+}
+
+void wt_transpose_launch(const long* src_ptrs, long* dst_ptrs, const int* Ks,
+                         const int* RCs, const int* tile_counts, int ntensors,
+                         int max_tiles, hipStream_t stream) {
+    dim3 grid((unsigned)max_tiles, (unsigned)ntensors);
+    hipLaunchKernelGGL(wt_transpose_kernel, grid, dim3(256), 0, stream,
+                       src_ptrs, dst_ptrs, Ks, RCs, tile_counts);
+}
+
+}  // namespace fda
+
+namespace fda {
+
+typedef __attribute__((ext_vector_type(4))) short short4_;
+
+// ---- wgrad: dw[k][rs][c] = sum_m dy[m][k] * x_gather[m][c] ----------------
+// GEMM with the reduction along the M (pixel) axis — both operands are
+// [m][channel] in memory, so the MFMA fragments (which want 8 elements
+// along the reduction per lane) are served by gfx950's ds_read_tr16_b64
+// hardware transpose-read from an "m4-grouped" LDS image:
+//   image element (m, ch) at  kb*1024 + (ch&15) + (m&3)*16 + (m>>2)*64
+//   (kb = ch/16); built directly by global_load_lds (16-B chunks are
+//   8-channel runs of one m row); read back with per-16-lane-group window
+//   addressing: lane addr = kb*2048B + (ks*8 + (lane>>4)*2)*128B +
+//   (lane&15)*8B, second half of the fragment at immediate offset +128B.
+// Each block owns a 64(k) x 64(c) output tile for one (r,s) tap and a
+// 4096-pixel M-chunk; chunks accumulate into an fp32 workspace with
+// atomicAdd (K*RS*C elements, cast to bf16 by the host wrapper).
+constexpr int WG_BM = 64;      // m per K-step
+constexpr int WG_MCH = 4096;   // pixels per block (chunk)
+constexpr int WG_TILE_ELEMS = WG_BM * 64;   // one operand tile
+
+__global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
+    const unsigned short* __restrict__ dy,   // [M][K] (NHWC out grad)
+    const unsigned short* __restrict__ x,    // [N,H,W,C]
+    float* __restrict__ ws,                  // [K][RS*C] fp32, pre-zeroed
+    int N, int H, int W, int C, int K, int P, int Q,
+    int R, int S, int sy, int sx, int py, int px, int nch) {
+    const int rs = blockIdx.z / nch;
+    const int chunk = blockIdx.z % nch;
+    const int r = rs / S, s = rs % S;
+    const int k0 = blockIdx.x * 64;
+    const int c0 = blockIdx.y * 64;
+    const long M = (long)N * P * Q;
+    const long mb0 = (long)chunk * WG_MCH;
+    const long mend = (mb0 + WG_MCH < M) ? mb0 + WG_MCH : M;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int wk = wid >> 1;          // 2x2 waves over [64k][64c]
+    const int wc = wid & 1;
+
+    extern __shared__ unsigned short lds[];   // [2][2*WG_TILE_ELEMS]
+
+    // staging decomposition: glds instr covers 1 KiB = 8 elems/lane;
+    // per wave per tile: 2 instrs. lane -> (kb, m_local, ch8):
+    //   e = (global lds offset)/8 elems... within tile: e8 = ln*8
+    //   kb = e8/1024; rr = e8%1024; m = (rr/64)*4 + (rr%64)/16;
+    //   ch = kb*16 + (rr%16)
+    const floatx4 zero4 = {0.f, 0.f, 0.f, 0.f};
+    floatx4 acc[2][2] = {{zero4, zero4}, {zero4, zero4}};
+
+    auto stage = [&](int buf, long mb) {
+        unsigned short* base = lds + buf * 2 * WG_TILE_ELEMS;
+        #pragma unroll
+        for (int t = 0; t < 2; ++t) {      // 0: dy tile, 1: x tile
+            #pragma unroll
+            for (int i = 0; i < 2; ++i) {
+                const int ln = (wid * 2 + i) * 64 + lane;  // 0..255 chunk id
+                const int e8 = ln * 8;
+                const int kb = e8 >> 10;
+                const int rr = e8 & 1023;
+                const int ml = ((rr >> 6) << 2) + ((rr & 63) >> 4);
+                const int ch = (kb << 4) + (rr & 15);
+                const long m = mb + ml;
+                const unsigned short* sp = conv_zero16;
+                if (m < mend) {
+                    if (t == 0) {
+                        sp = dy + m * K + k0 + ch;
+                    } else {
+                        const int qq = (int)(m % Q);
+                        const int pp = (int)((m / Q) % P);
+                        const int n = (int)(m / ((long)Q * P));
+                        const int hh = pp * sy - py + r;
+                        const int ww = qq * sx - px + s;
+                        if ((unsigned)hh < (unsigned)H &&
+                            (unsigned)ww < (unsigned)W)
+                            sp = x + (((long)n * H + hh) * W + ww) * C + c0 + ch;
+                    }
+                }
+                FDA_GLDS16(sp, base + t * WG_TILE_ELEMS + (wid * 2 + i) * 512);
+            }
+        }
+    };
+    // NOTE: the glds lds destination advances lane*16B from the wave base;
+    // chunk id ln above must equal (wave base + lane), i.e. instr i of wave
+    // w writes chunks [(w*2+i)*64 .. +64). ln matches that. Each 16-B chunk
+    // holds {m=ml, ch8 run}, which is exactly the image layout: chunk c ->
+    // elems [c*8, c*8+8).
+
+    // tr-read base addresses (bytes within one operand tile) per fragment:
+    //   frag(op, fi, ks): kb = (wv*32 + fi*16)/16, window = ks*8+(l>>4)*2
+    const int l15 = lane & 15, lg = lane >> 4;
+
+    const int nsteps = (int)((mend - mb0 + WG_BM - 1) / WG_BM);
+    if (nsteps > 0) stage(0, mb0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    for (int it = 0; it < nsteps; ++it) {
+        if (it + 1 < nsteps) stage((it + 1) & 1, mb0 + (long)(it + 1) * WG_BM);
+        const unsigned short* buf = lds + (it & 1) * 2 * WG_TILE_ELEMS;
+        short4_ a[2][2][2], b[2][2][2];   // [fi][ks][half]
+        #pragma unroll
+        for (int fi = 0; fi < 2; ++fi)
+            #pragma unroll
+            for (int ks = 0; ks < 2; ++ks) {
+                const int kb_a = wk * 2 + fi;
+                const int kb_b = wc * 2 + fi;
+                const unsigned short* pa =
+                    buf + kb_a * 1024 + (ks * 8 + lg * 2) * 64 + l15 * 4;
+                const unsigned short* pb =
+                    buf + WG_TILE_ELEMS + kb_b * 1024 +
+                    (ks * 8 + lg * 2) * 64 + l15 * 4;
+                a[fi][ks][0] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (__attribute__((address_space(3))) short4_*)pa);
+                a[fi][ks][1] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (__attribute__((address_space(3))) short4_*)(pa + 64));
+                b[fi][ks][0] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (__attribute__((address_space(3))) short4_*)pb);
+                b[fi][ks][1] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (__attribute__((address_space(3))) short4_*)(pb + 64));
+            }
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+            #pragma unroll
+            for (int ki = 0; ki < 2; ++ki)
+                #pragma unroll
+                for (int ci = 0; ci < 2; ++ci) {
+                    short8 af, bf;
+                    #pragma unroll
+                    for (int e = 0; e < 4; ++e) {
+                        af[e] = a[ki][ks][0][e];
+                        af[e + 4] = a[ki][ks][1][e];
+                        bf[e] = b[ci][ks][0][e];
+                        bf[e + 4] = b[ci][ks][1][e];
+                    }
+                    acc[ki][ci] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af, bf, acc[ki][ci], 0, 0, 0);
+                }
+        __builtin_amdgcn_s_setprio(0);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+    }
+
+    // epilogue: out[i=k][j=c]; C/D map col=lane&15, row=(lane>>4)*4+jj
+    const int fcol = lane & 15, frow0 = (lane >> 4) * 4;
+    const long RSC = (long)R * S * C;
+    #pragma unroll
+    for (int ki = 0; ki < 2; ++ki)
+        #pragma unroll
+        for (int ci = 0; ci < 2; ++ci)
+            #pragma unroll
+            for (int jj = 0; jj < 4; ++jj) {
+                const int kk = k0 + wk * 32 + ki * 16 + frow0 + jj;
+                const int cc = c0 + wc * 32 + ci * 16 + fcol;
+                atomicAdd(&ws[kk * RSC + (long)rs * C + cc],
+                          acc[ki][ci][jj]);
+            }
+}
+
+void conv_wgrad_launch(const void* dy, const void* x, float* ws,
+                       int N, int H, int W, int C, int K, int P, int Q,
+                       int R, int S, int sy, int sx, int py, int px,
+                       hipStream_t stream) {
+    const long M = (long)N * P * Q;
+    const int nch = (int)((M + WG_MCH - 1) / WG_MCH);
+    dim3 grid((unsigned)(K / 64), (unsigned)(C / 64),
+              (unsigned)(R * S * nch));
+    const size_t shmem = 2 * 2 * WG_TILE_ELEMS * sizeof(unsigned short);
+    hipLaunchKernelGGL(conv_wgrad_kernel, grid, dim3(256), shmem, stream,
+                       (const unsigned short*)dy, (const unsigned short*)x,
+                       ws, N, H, W, C, K, P, Q, R, S, sy, sx, py, px, nch);
+}
+
+}  // namespace fda

@@ -71,6 +71,18 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int R, int S, int sy, int sx, int py, int px,
                        bool dgrad, hipStream_t stream);
 
+// wgrad: ws[K][RS*C] fp32 (pre-zeroed) += dy^T @ im2col(x), atomic chunks
+void conv_wgrad_launch(const void* dy, const void* x, float* ws,
+                       int N, int H, int W, int C, int K, int P, int Q,
+                       int R, int S, int sy, int sx, int py, int px,
+                       hipStream_t stream);
+
+// batched conv-weight transpose: w[k][rc] -> wt[rc][k] for all tensors in
+// one launch (device pointer arrays)
+void wt_transpose_launch(const long* src_ptrs, long* dst_ptrs, const int* Ks,
+                         const int* RCs, const int* tile_counts, int ntensors,
+                         int max_tiles, hipStream_t stream);
+
 // flat fused optimizers. P: param dtype; M/V/S fp32; G param dtype.
 void sgd_step_launch(void* P, const void* G, float* M, float* V, int64_t n,
                      float lr, float mom, float wd, bool nesterov,

@@ -40,6 +40,83 @@ def _native_supported(x: torch.Tensor, w: torch.Tensor, stride, padding,
     return load_native() is not None
 
 
+class _WtArena:
+    """Per-device cache of transposed conv weights wt[rs*C+c][k] for dgrad.
+
+    All registered weights are transposed in ONE kernel launch per training
+    step (wt_transpose_batch) instead of 36 per-layer permutes. Freshness:
+    a marker bumped by the fused optimizers (ops/fused_optim.py write params
+    through raw kernels, invisible to autograd) plus the sum of the weights'
+    `_version` counters (covers torch.optim in-place updates).
+    """
+
+    def __init__(self, device):
+        self.device = device
+        self.weights = []               # [(weight, K, RC)]
+        self.slices = {}                # id(weight) -> (offset, RC, K)
+        self.arena = None
+        self.meta = None
+        self.fresh_key = None
+
+    def register(self, weight):
+        if id(weight) in self.slices:
+            return
+        K, Cin, R, S = weight.shape
+        RC = R * S * Cin
+        self.weights.append((weight, K, RC))
+        self.slices[id(weight)] = (None, RC, K)
+        self.arena = None               # rebuild on next get
+
+    def _build(self):
+        dev = self.device
+        total = sum(K * RC for (_, K, RC) in self.weights)
+        self.arena = torch.empty(total, dtype=torch.bfloat16, device=dev)
+        off = 0
+        srcs, dsts, Ks, RCs, tiles = [], [], [], [], []
+        for (w, K, RC) in self.weights:
+            self.slices[id(w)] = (off, RC, K)
+            srcs.append(w.data_ptr())
+            dsts.append(self.arena.data_ptr() + off * 2)
+            Ks.append(K)
+            RCs.append(RC)
+            tiles.append((K // 64) * (RC // 64))
+            off += K * RC
+        mk = lambda v, dt: torch.tensor(v, dtype=dt, device=dev)
+        self.meta = (mk(srcs, torch.long), mk(dsts, torch.long),
+                     mk(Ks, torch.int32), mk(RCs, torch.int32),
+                     mk(tiles, torch.int32), max(tiles))
+
+    def get(self, weight):
+        key = (_WT_MARKER[0], sum(w._version for (w, _, _) in self.weights))
+        if self.arena is None:
+            self._build()
+            self.fresh_key = None
+        if self.fresh_key != key:
+            C = require_native("wt_transpose_batch")
+            s, d, k, rc, t, mt = self.meta
+            C.wt_transpose_batch(s, d, k, rc, t, mt)
+            self.fresh_key = key
+        off, RC, K = self.slices[id(weight)]
+        return self.arena[off : off + RC * K].view(RC, K)
+
+
+_WT_MARKER = [0]
+_ARENAS: dict = {}
+
+
+def bump_conv_wt_marker():
+    """Invalidate cached transposed weights. Called by the fused optimizers
+    after each raw-kernel parameter update."""
+    _WT_MARKER[0] += 1
+
+
+def _arena_for(weight) -> _WtArena:
+    dev = weight.device
+    if dev not in _ARENAS:
+        _ARENAS[dev] = _WtArena(dev)
+    return _ARENAS[dev]
+
+
 class _FdaConv2d(torch.autograd.Function):
     """Forward + input-grad on the native implicit-GEMM kernels; weight-grad
     via the library (aten convolution_backward with weight-only mask)."""
@@ -53,13 +130,15 @@ class _FdaConv2d(torch.autograd.Function):
         py, px = padding
         y = C.conv_igemm_fwd(xc, wc, sy, sx, py, px)
         ctx.save_for_backward(xc, wc)
-        ctx.conf = (stride, padding)
+        ctx.conf = (stride, padding, weight is wc)
+        if weight is wc:  # already channels_last: cacheable by identity
+            _arena_for(wc).register(wc)
         return y
 
     @staticmethod
     def backward(ctx, gy):
         x, w = ctx.saved_tensors
-        stride, padding = ctx.conf
+        stride, padding, cacheable = ctx.conf
         sy, sx = stride
         py, px = padding
         gy = gy.contiguous(memory_format=torch.channels_last)
@@ -67,14 +146,26 @@ class _FdaConv2d(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             C = require_native("conv_igemm_dgrad")
             K, Cin, R, S = w.shape
-            # wt[(r*S+s)*C + c][k]: k-contiguous rows for the staged B tile
-            wt = w.permute(2, 3, 1, 0).reshape(R * S * Cin, K).contiguous()
+            if cacheable:
+                wt = _arena_for(w).get(w)
+            else:
+                # wt[(r*S+s)*C + c][k]: k-contiguous rows for the B tile
+                wt = w.permute(2, 3, 1, 0).reshape(R * S * Cin, K).contiguous()
             dx = C.conv_igemm_dgrad(gy, wt, Cin, x.shape[2], x.shape[3],
                                     R, S, sy, sx, py, px)
         if ctx.needs_input_grad[1]:
-            dw = torch.ops.aten.convolution_backward(
-                gy, x, w, None, list(stride), list(padding), [1, 1],
-                False, [0, 0], 1, [False, True, False])[1]
+            K, Cin, R, S = w.shape
+            if os.environ.get("FLUXDIST_WGRAD", "") != "miopen":
+                C = require_native("conv_igemm_wgrad")
+                ws = C.conv_igemm_wgrad(gy, x, R, S, sy, sx, py, px)
+                dw = torch.empty_like(w)
+                # ws [K][R*S*C] fp32 is exactly the channels_last weight
+                # memory order [K][R][S][C]
+                dw.copy_(ws.view(K, R, S, Cin).permute(0, 3, 1, 2))
+            else:
+                dw = torch.ops.aten.convolution_backward(
+                    gy, x, w, None, list(stride), list(padding), [1, 1],
+                    False, [0, 0], 1, [False, True, False])[1]
         return dx, dw, None, None
 
 

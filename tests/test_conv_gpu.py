@@ -103,3 +103,25 @@ def test_model_uses_native_conv():
     m = build_model("resnet34")
     fda = [mod for mod in m.modules() if isinstance(mod, FdaConv2d)]
     assert len(fda) >= 35, f"expected >=35 FdaConv2d body convs, got {len(fda)}"
+
+
+@pytest.mark.parametrize("shape", SHAPES)
+def test_conv_wgrad_matches_fp32(shape):
+    n, c, h, w, k, r, s = shape
+    x, wt = _mk(n, c, h, w, k, r, seed=3)
+    pad = r // 2
+    w32 = wt.clone().requires_grad_(True)
+    ref = torch.nn.functional.conv2d(x, w32, stride=s, padding=pad)
+    gy = torch.randn_like(ref)
+    ref.backward(gy)
+    dw_ref = w32.grad
+
+    C = require_native("conv_igemm_wgrad")
+    gyb = gy.bfloat16().contiguous(memory_format=torch.channels_last)
+    xb = x.bfloat16().contiguous(memory_format=torch.channels_last)
+    ws = C.conv_igemm_wgrad(gyb, xb, r, r, s, s, pad, pad)
+    dw = ws.view(k, r, r, c).permute(0, 3, 1, 2)
+    err = (dw - dw_ref).abs().max().item()
+    scale = dw_ref.abs().max().item()
+    # reduction over N*P*Q in bf16 products, fp32 accumulate
+    assert err < 0.02 * max(scale, 1.0), f"{shape}: err={err} scale={scale}"

@@ -60,8 +60,9 @@ def main():
 
     print(f"# batch={n}  times in us, eff TF = 2*M*N*K / t")
     print(f"{'shape':34s} {'fda_fwd':>9s} {'lib_fwd':>9s} {'fda_TF':>7s} {'lib_TF':>7s}"
-          f" | {'fda_dg':>9s} {'lib_dg':>9s}")
+          f" | {'fda_dg':>9s} {'lib_dg':>9s} | {'fda_wg':>9s} {'lib_wg':>9s}")
     tot_fda_f = tot_lib_f = tot_fda_d = tot_lib_d = 0.0
+    tot_fda_w = tot_lib_w = 0.0
     shapes = SHAPES if args.only < 0 else [SHAPES[args.only]]
     for (c, h, w, k, r, s, cnt) in shapes:
         pad = r // 2
@@ -87,18 +88,29 @@ def main():
                 gy, xg, wt, None, [s, s], [pad, pad], [1, 1], False, [0, 0],
                 1, [True, False, False])[0]
         t_ld = timeit(lib_dgrad, args.iters) if not args.fda_only else float("inf")
+        t_fw = timeit(lambda: C_.conv_igemm_wgrad(gy, x, r, r, s, s, pad, pad),
+                      args.iters)
+
+        def lib_wgrad():
+            return torch.ops.aten.convolution_backward(
+                gy, xg, wt, None, [s, s], [pad, pad], [1, 1], False, [0, 0],
+                1, [False, True, False])[1]
+        t_lw = timeit(lib_wgrad, args.iters) if not args.fda_only else float("inf")
 
         name = f"{c}x{h}x{w} k{k} {r}x{r} s{s} x{cnt}"
         print(f"{name:34s} {t_ff*1e6:9.1f} {t_lf*1e6:9.1f} "
               f"{flops/t_ff/1e12:7.1f} {flops/t_lf/1e12:7.1f} | "
-              f"{t_fd*1e6:9.1f} {t_ld*1e6:9.1f}")
+              f"{t_fd*1e6:9.1f} {t_ld*1e6:9.1f} | "
+              f"{t_fw*1e6:9.1f} {t_lw*1e6:9.1f}")
         tot_fda_f += t_ff * cnt
         tot_lib_f += t_lf * cnt
         tot_fda_d += t_fd * cnt
         tot_lib_d += t_ld * cnt
+        tot_fda_w += t_fw * cnt
+        tot_lib_w += t_lw * cnt
     print(f"{'TOTAL (weighted by layer count)':34s} {tot_fda_f*1e6:9.1f} "
           f"{tot_lib_f*1e6:9.1f} {'':7s} {'':7s} | {tot_fda_d*1e6:9.1f} "
-          f"{tot_lib_d*1e6:9.1f}")
+          f"{tot_lib_d*1e6:9.1f} | {tot_fda_w*1e6:9.1f} {tot_lib_w*1e6:9.1f}")
 
 
 if __name__ == "__main__":
