@@ -377,7 +377,7 @@ typedef __attribute__((ext_vector_type(4))) short short4_;
 // 4096-pixel M-chunk; chunks accumulate into an fp32 workspace with
 // atomicAdd (K*RS*C elements, cast to bf16 by the host wrapper).
 constexpr int WG_BM = 64;      // m per K-step
-constexpr int WG_MCH = 4096;   // pixels per block (chunk)
+constexpr int WG_MCH = 2048;   // pixels per block (chunk)
 constexpr int WG_TILE_ELEMS = WG_BM * 64;   // one operand tile
 
 __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
@@ -401,7 +401,7 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
     const int wk = wid >> 1;          // 2x2 waves over [64k][64c]
     const int wc = wid & 1;
 
-    extern __shared__ unsigned short lds[];   // [2][2*WG_TILE_ELEMS]
+    extern __shared__ unsigned short lds[];   // [3][2*WG_TILE_ELEMS]
 
     // staging decomposition: glds instr covers 1 KiB = 8 elems/lane;
     // per wave per tile: 2 instrs. lane -> (kb, m_local, ch8):
@@ -453,13 +453,20 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
     //   frag(op, fi, ks): kb = (wv*32 + fi*16)/16, window = ks*8+(l>>4)*2
     const int l15 = lane & 15, lg = lane >> 4;
 
+    // 3-buffer ring, counted vmcnt: tile t+2's DMA stays in flight across
+    // the barrier while tile t computes (4 glds per wave per tile-pair).
     const int nsteps = (int)((mend - mb0 + WG_BM - 1) / WG_BM);
     if (nsteps > 0) stage(0, mb0);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
+    if (nsteps > 1) stage(1, mb0 + WG_BM);
     for (int it = 0; it < nsteps; ++it) {
-        if (it + 1 < nsteps) stage((it + 1) & 1, mb0 + (long)(it + 1) * WG_BM);
-        const unsigned short* buf = lds + (it & 1) * 2 * WG_TILE_ELEMS;
+        if (it + 1 < nsteps)
+            asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (it + 2 < nsteps)
+            stage((it + 2) % 3, mb0 + (long)(it + 2) * WG_BM);
+        const unsigned short* buf = lds + (it % 3) * 2 * WG_TILE_ELEMS;
         short4_ a[2][2][2], b[2][2][2];   // [fi][ks][half]
         #pragma unroll
         for (int fi = 0; fi < 2; ++fi)
@@ -500,7 +507,6 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
                         af, bf, acc[ki][ci], 0, 0, 0);
                 }
         __builtin_amdgcn_s_setprio(0);
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
     }
 
@@ -528,7 +534,7 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
     const int nch = (int)((M + WG_MCH - 1) / WG_MCH);
     dim3 grid((unsigned)(K / 64), (unsigned)(C / 64),
               (unsigned)(R * S * nch));
-    const size_t shmem = 2 * 2 * WG_TILE_ELEMS * sizeof(unsigned short);
+    const size_t shmem = 3 * 2 * WG_TILE_ELEMS * sizeof(unsigned short);
     hipLaunchKernelGGL(conv_wgrad_kernel, grid, dim3(256), shmem, stream,
                        (const unsigned short*)dy, (const unsigned short*)x,
                        ws, N, H, W, C, K, P, Q, R, S, sy, sx, py, px, nch);
