@@ -158,10 +158,10 @@ class _FdaConv2d(torch.autograd.Function):
             if os.environ.get("FLUXDIST_WGRAD", "") != "miopen":
                 C = require_native("conv_igemm_wgrad")
                 ws = C.conv_igemm_wgrad(gy, x, R, S, sy, sx, py, px)
-                dw = torch.empty_like(w)
                 # ws [K][R*S*C] fp32 is exactly the channels_last weight
-                # memory order [K][R][S][C]
-                dw.copy_(ws.view(K, R, S, Cin).permute(0, 3, 1, 2))
+                # memory order [K][R][S][C]: one flat cast, zero-copy view
+                dw = (ws.to(torch.bfloat16).view(K, R, S, Cin)
+                      .permute(0, 3, 1, 2))
             else:
                 dw = torch.ops.aten.convolution_backward(
                     gy, x, w, None, list(stride), list(padding), [1, 1],

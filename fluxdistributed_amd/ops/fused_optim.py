@@ -23,6 +23,14 @@ import torch
 from .native import require_native
 
 
+def _bump_wt_marker():
+    # raw-kernel param writes are invisible to autograd version counters;
+    # invalidate the dgrad transposed-weight arena (ops/conv.py)
+    from .conv import bump_conv_wt_marker
+
+    bump_conv_wt_marker()
+
+
 class _FlatGroup:
     """All parameters of one (device, dtype) flattened into shared storage."""
 
@@ -144,6 +152,7 @@ class FusedSGDMomentum(_FlatOptimizer):
                 C = require_native("fused_sgd")
                 C.sgd_step(g.P, g.G, g.master if g.master is not None else g.P,
                            g.V, lr, mom, wd, nesterov)
+                _bump_wt_marker()
             else:
                 master = g.master if g.master is not None else g.P
                 grad = g.G.float()
@@ -182,6 +191,7 @@ class FusedAdam(_FlatOptimizer):
                 C = require_native("fused_adam")
                 C.adam_step(g.P, g.G, g.master if g.master is not None else g.P,
                             g.V, g.S, lr, b1, b2, eps, wd, bc1, bc2)
+                _bump_wt_marker()
             else:
                 master = g.master if g.master is not None else g.P
                 grad = g.G.float()
