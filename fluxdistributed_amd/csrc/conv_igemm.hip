@@ -411,36 +411,66 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
     const floatx4 zero4 = {0.f, 0.f, 0.f, 0.f};
     floatx4 acc[2][2] = {{zero4, zero4}, {zero4, zero4}};
 
-    auto stage = [&](int buf, long mb) {
+    // Per-lane staging coordinates, advanced INCREMENTALLY: stage() is
+    // called with strictly sequential m-bases (prologue 0,1 then it+2), so
+    // each lane tracks its pixel (n,p,q) with constant-delta carries — no
+    // per-iteration division (those were ~3x the MFMA issue time here).
+    int st_q[2], st_p[2], st_n[2], st_ch[2], st_ml[2];
+    long st_dyoff[2];
+    long st_m[2];
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {
+        const int ln = (wid * 2 + i) * 64 + lane;
+        const int e8 = ln * 8;
+        const int kb = e8 >> 10;
+        const int rr = e8 & 1023;
+        const int ml = ((rr >> 6) << 2) + ((rr & 63) >> 4);
+        const int ch = (kb << 4) + (rr & 15);
+        st_ml[i] = ml;
+        st_ch[i] = ch;
+        const long m = mb0 + ml;
+        st_m[i] = m;
+        st_q[i] = (int)(m % Q);
+        st_p[i] = (int)((m / Q) % P);
+        st_n[i] = (int)(m / ((long)Q * P));
+        st_dyoff[i] = m * K + k0 + ch;
+    }
+    const int dQ = WG_BM % Q, dP = (WG_BM / Q) % P, dN0 = WG_BM / (Q * P);
+    const long dDY = (long)WG_BM * K;
+
+    auto stage = [&](int buf) {
         unsigned short* base = lds + buf * 2 * WG_TILE_ELEMS;
         #pragma unroll
-        for (int t = 0; t < 2; ++t) {      // 0: dy tile, 1: x tile
-            #pragma unroll
-            for (int i = 0; i < 2; ++i) {
-                const int ln = (wid * 2 + i) * 64 + lane;  // 0..255 chunk id
-                const int e8 = ln * 8;
-                const int kb = e8 >> 10;
-                const int rr = e8 & 1023;
-                const int ml = ((rr >> 6) << 2) + ((rr & 63) >> 4);
-                const int ch = (kb << 4) + (rr & 15);
-                const long m = mb + ml;
-                const unsigned short* sp = conv_zero16;
-                if (m < mend) {
-                    if (t == 0) {
-                        sp = dy + m * K + k0 + ch;
-                    } else {
-                        const int qq = (int)(m % Q);
-                        const int pp = (int)((m / Q) % P);
-                        const int n = (int)(m / ((long)Q * P));
-                        const int hh = pp * sy - py + r;
-                        const int ww = qq * sx - px + s;
-                        if ((unsigned)hh < (unsigned)H &&
-                            (unsigned)ww < (unsigned)W)
-                            sp = x + (((long)n * H + hh) * W + ww) * C + c0 + ch;
-                    }
-                }
-                FDA_GLDS16(sp, base + t * WG_TILE_ELEMS + (wid * 2 + i) * 512);
+        for (int i = 0; i < 2; ++i) {
+            const unsigned short* sp =
+                (st_m[i] < mend) ? dy + st_dyoff[i] : conv_zero16;
+            FDA_GLDS16(sp, base + (wid * 2 + i) * 512);
+        }
+        #pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            const unsigned short* sp = conv_zero16;
+            if (st_m[i] < mend) {
+                const int hh = st_p[i] * sy - py + r;
+                const int ww = st_q[i] * sx - px + s;
+                if ((unsigned)hh < (unsigned)H && (unsigned)ww < (unsigned)W)
+                    sp = x + (((long)st_n[i] * H + hh) * W + ww) * C +
+                         c0 + st_ch[i];
             }
+            FDA_GLDS16(sp, base + WG_TILE_ELEMS + (wid * 2 + i) * 512);
+        }
+        // advance 64 pixels (bounded carries; dP < 2P, so p needs at most
+        // two conditional wraps after the q carry)
+        #pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            st_m[i] += WG_BM;
+            st_dyoff[i] += dDY;
+            int q = st_q[i] + dQ;
+            int p = st_p[i] + dP;
+            int n = st_n[i] + dN0;
+            if (q >= Q) { q -= Q; ++p; }
+            if (p >= P) { p -= P; ++n; }
+            if (p >= P) { p -= P; ++n; }
+            st_q[i] = q; st_p[i] = p; st_n[i] = n;
         }
     };
     // NOTE: the glds lds destination advances lane*16B from the wave base;
@@ -456,8 +486,8 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
     // 3-buffer ring, counted vmcnt: tile t+2's DMA stays in flight across
     // the barrier while tile t computes (4 glds per wave per tile-pair).
     const int nsteps = (int)((mend - mb0 + WG_BM - 1) / WG_BM);
-    if (nsteps > 0) stage(0, mb0);
-    if (nsteps > 1) stage(1, mb0 + WG_BM);
+    if (nsteps > 0) stage(0);
+    if (nsteps > 1) stage(1);
     for (int it = 0; it < nsteps; ++it) {
         if (it + 1 < nsteps)
             asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
@@ -465,7 +495,7 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
         if (it + 2 < nsteps)
-            stage((it + 2) % 3, mb0 + (long)(it + 2) * WG_BM);
+            stage((it + 2) % 3);
         const unsigned short* buf = lds + (it % 3) * 2 * WG_TILE_ELEMS;
         short4_ a[2][2][2], b[2][2][2];   // [fi][ks][half]
         #pragma unroll
