@@ -373,24 +373,28 @@ typedef __attribute__((ext_vector_type(4))) short short4_;
 //   8-channel runs of one m row); read back with per-16-lane-group window
 //   addressing: lane addr = kb*2048B + (ks*8 + (lane>>4)*2)*128B +
 //   (lane&15)*8B, second half of the fragment at immediate offset +128B.
-// Each block owns a 64(k) x 64(c) output tile for one (r,s) tap and a
-// 4096-pixel M-chunk; chunks accumulate into an fp32 workspace with
-// atomicAdd (K*RS*C elements, cast to bf16 by the host wrapper).
+// Each block owns a (32*FT)k x (32*FT)c output tile (FT = fragments per
+// wave axis: 2 -> 64x64 tile / 8 MFMAs per K-step, 4 -> 128x128 tile /
+// 32 MFMAs — chosen by channel divisibility; the bigger tile quadruples
+// the MFMA-per-glds ratio) for one (r,s) tap and a 2048-pixel M-chunk;
+// chunks accumulate into an fp32 workspace with atomicAdd.
 constexpr int WG_BM = 64;      // m per K-step
 constexpr int WG_MCH = 2048;   // pixels per block (chunk)
-constexpr int WG_TILE_ELEMS = WG_BM * 64;   // one operand tile
 
-__global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
+template <int FT>
+__global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
     const unsigned short* __restrict__ dy,   // [M][K] (NHWC out grad)
     const unsigned short* __restrict__ x,    // [N,H,W,C]
     float* __restrict__ ws,                  // [K][RS*C] fp32, pre-zeroed
     int N, int H, int W, int C, int K, int P, int Q,
     int R, int S, int sy, int sx, int py, int px, int nch) {
+    constexpr int TCH = 32 * FT;             // tile channels per operand
+    constexpr int TILE_ELEMS = WG_BM * TCH;  // one operand tile
     const int rs = blockIdx.z / nch;
     const int chunk = blockIdx.z % nch;
     const int r = rs / S, s = rs % S;
-    const int k0 = blockIdx.x * 64;
-    const int c0 = blockIdx.y * 64;
+    const int k0 = blockIdx.x * TCH;
+    const int c0 = blockIdx.y * TCH;
     const long M = (long)N * P * Q;
     const long mb0 = (long)chunk * WG_MCH;
     const long mend = (mb0 + WG_MCH < M) ? mb0 + WG_MCH : M;
@@ -398,35 +402,33 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wid = tid >> 6;
-    const int wk = wid >> 1;          // 2x2 waves over [64k][64c]
+    const int wk = wid >> 1;          // 2x2 waves over [TCH k][TCH c]
     const int wc = wid & 1;
 
-    extern __shared__ unsigned short lds[];   // [3][2*WG_TILE_ELEMS]
+    extern __shared__ unsigned short lds[];   // [ring][2*TILE_ELEMS]
 
-    // staging decomposition: glds instr covers 1 KiB = 8 elems/lane;
-    // per wave per tile: 2 instrs. lane -> (kb, m_local, ch8):
-    //   e = (global lds offset)/8 elems... within tile: e8 = ln*8
-    //   kb = e8/1024; rr = e8%1024; m = (rr/64)*4 + (rr%64)/16;
-    //   ch = kb*16 + (rr%16)
     const floatx4 zero4 = {0.f, 0.f, 0.f, 0.f};
-    floatx4 acc[2][2] = {{zero4, zero4}, {zero4, zero4}};
+    floatx4 acc[FT][FT];
+    #pragma unroll
+    for (int ki = 0; ki < FT; ++ki)
+        #pragma unroll
+        for (int ci = 0; ci < FT; ++ci) acc[ki][ci] = zero4;
 
     // Per-lane staging coordinates, advanced INCREMENTALLY: stage() is
     // called with strictly sequential m-bases (prologue 0,1 then it+2), so
     // each lane tracks its pixel (n,p,q) with constant-delta carries — no
     // per-iteration division (those were ~3x the MFMA issue time here).
-    int st_q[2], st_p[2], st_n[2], st_ch[2], st_ml[2];
-    long st_dyoff[2];
-    long st_m[2];
+    int st_q[FT], st_p[FT], st_n[FT], st_ch[FT];
+    long st_dyoff[FT];
+    long st_m[FT];
     #pragma unroll
-    for (int i = 0; i < 2; ++i) {
-        const int ln = (wid * 2 + i) * 64 + lane;
+    for (int i = 0; i < FT; ++i) {
+        const int ln = (wid * FT + i) * 64 + lane;
         const int e8 = ln * 8;
         const int kb = e8 >> 10;
         const int rr = e8 & 1023;
         const int ml = ((rr >> 6) << 2) + ((rr & 63) >> 4);
         const int ch = (kb << 4) + (rr & 15);
-        st_ml[i] = ml;
         st_ch[i] = ch;
         const long m = mb0 + ml;
         st_m[i] = m;
@@ -439,15 +441,15 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
     const long dDY = (long)WG_BM * K;
 
     auto stage = [&](int buf) {
-        unsigned short* base = lds + buf * 2 * WG_TILE_ELEMS;
+        unsigned short* base = lds + buf * 2 * TILE_ELEMS;
         #pragma unroll
-        for (int i = 0; i < 2; ++i) {
+        for (int i = 0; i < FT; ++i) {
             const unsigned short* sp =
                 (st_m[i] < mend) ? dy + st_dyoff[i] : conv_zero16;
-            FDA_GLDS16(sp, base + (wid * 2 + i) * 512);
+            FDA_GLDS16(sp, base + (wid * FT + i) * 512);
         }
         #pragma unroll
-        for (int i = 0; i < 2; ++i) {
+        for (int i = 0; i < FT; ++i) {
             const unsigned short* sp = conv_zero16;
             if (st_m[i] < mend) {
                 const int hh = st_p[i] * sy - py + r;
@@ -456,12 +458,12 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
                     sp = x + (((long)st_n[i] * H + hh) * W + ww) * C +
                          c0 + st_ch[i];
             }
-            FDA_GLDS16(sp, base + WG_TILE_ELEMS + (wid * 2 + i) * 512);
+            FDA_GLDS16(sp, base + TILE_ELEMS + (wid * FT + i) * 512);
         }
-        // advance 64 pixels (bounded carries; dP < 2P, so p needs at most
+        // advance 64 pixels (bounded carries; dP < P, so p needs at most
         // two conditional wraps after the q carry)
         #pragma unroll
-        for (int i = 0; i < 2; ++i) {
+        for (int i = 0; i < FT; ++i) {
             st_m[i] += WG_BM;
             st_dyoff[i] += dDY;
             int q = st_q[i] + dQ;
@@ -473,41 +475,38 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
             st_q[i] = q; st_p[i] = p; st_n[i] = n;
         }
     };
-    // NOTE: the glds lds destination advances lane*16B from the wave base;
-    // chunk id ln above must equal (wave base + lane), i.e. instr i of wave
-    // w writes chunks [(w*2+i)*64 .. +64). ln matches that. Each 16-B chunk
-    // holds {m=ml, ch8 run}, which is exactly the image layout: chunk c ->
-    // elems [c*8, c*8+8).
 
-    // tr-read base addresses (bytes within one operand tile) per fragment:
-    //   frag(op, fi, ks): kb = (wv*32 + fi*16)/16, window = ks*8+(l>>4)*2
     const int l15 = lane & 15, lg = lane >> 4;
 
-    // 3-buffer ring, counted vmcnt: tile t+2's DMA stays in flight across
-    // the barrier while tile t computes (4 glds per wave per tile-pair).
+    // ring of RING tile-pairs, counted vmcnt: tile t+2's DMA stays in
+    // flight across the barrier while tile t computes (2*FT glds per wave
+    // per tile-pair). FT=4 uses a 2-deep ring (96 KiB would exceed 1
+    // block/CU headroom at 3).
+    constexpr int RING = (FT == 2) ? 3 : 2;
     const int nsteps = (int)((mend - mb0 + WG_BM - 1) / WG_BM);
     if (nsteps > 0) stage(0);
-    if (nsteps > 1) stage(1);
+    if (RING > 2 && nsteps > 1) stage(1);
     for (int it = 0; it < nsteps; ++it) {
-        if (it + 1 < nsteps)
-            asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-        else
+        if (RING > 2 && it + 1 < nsteps) {
+            if (FT == 2) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+            else         asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+        } else {
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
         __builtin_amdgcn_s_barrier();
-        if (it + 2 < nsteps)
-            stage((it + 2) % 3);
-        const unsigned short* buf = lds + (it % 3) * 2 * WG_TILE_ELEMS;
-        short4_ a[2][2][2], b[2][2][2];   // [fi][ks][half]
+        if (it + RING - 1 < nsteps) stage((it + RING - 1) % RING);
+        const unsigned short* buf = lds + (it % RING) * 2 * TILE_ELEMS;
+        short4_ a[FT][2][2], b[FT][2][2];   // [fi][ks][half]
         #pragma unroll
-        for (int fi = 0; fi < 2; ++fi)
+        for (int fi = 0; fi < FT; ++fi)
             #pragma unroll
             for (int ks = 0; ks < 2; ++ks) {
-                const int kb_a = wk * 2 + fi;
-                const int kb_b = wc * 2 + fi;
+                const int kb_a = wk * FT + fi;
+                const int kb_b = wc * FT + fi;
                 const unsigned short* pa =
                     buf + kb_a * 1024 + (ks * 8 + lg * 2) * 64 + l15 * 4;
                 const unsigned short* pb =
-                    buf + WG_TILE_ELEMS + kb_b * 1024 +
+                    buf + TILE_ELEMS + kb_b * 1024 +
                     (ks * 8 + lg * 2) * 64 + l15 * 4;
                 a[fi][ks][0] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
                     (__attribute__((address_space(3))) short4_*)pa);
@@ -522,9 +521,9 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
         #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
             #pragma unroll
-            for (int ki = 0; ki < 2; ++ki)
+            for (int ki = 0; ki < FT; ++ki)
                 #pragma unroll
-                for (int ci = 0; ci < 2; ++ci) {
+                for (int ci = 0; ci < FT; ++ci) {
                     short8 af, bf;
                     #pragma unroll
                     for (int e = 0; e < 4; ++e) {
@@ -537,6 +536,9 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
                         af, bf, acc[ki][ci], 0, 0, 0);
                 }
         __builtin_amdgcn_s_setprio(0);
+        if (RING == 2) {
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
         __builtin_amdgcn_s_barrier();
     }
 
@@ -544,13 +546,13 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(
     const int fcol = lane & 15, frow0 = (lane >> 4) * 4;
     const long RSC = (long)R * S * C;
     #pragma unroll
-    for (int ki = 0; ki < 2; ++ki)
+    for (int ki = 0; ki < FT; ++ki)
         #pragma unroll
-        for (int ci = 0; ci < 2; ++ci)
+        for (int ci = 0; ci < FT; ++ci)
             #pragma unroll
             for (int jj = 0; jj < 4; ++jj) {
-                const int kk = k0 + wk * 32 + ki * 16 + frow0 + jj;
-                const int cc = c0 + wc * 32 + ci * 16 + fcol;
+                const int kk = k0 + wk * 16 * FT + ki * 16 + frow0 + jj;
+                const int cc = c0 + wc * 16 * FT + ci * 16 + fcol;
                 atomicAdd(&ws[kk * RSC + (long)rs * C + cc],
                           acc[ki][ci][jj]);
             }
@@ -562,12 +564,23 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
                        hipStream_t stream) {
     const long M = (long)N * P * Q;
     const int nch = (int)((M + WG_MCH - 1) / WG_MCH);
-    dim3 grid((unsigned)(K / 64), (unsigned)(C / 64),
-              (unsigned)(R * S * nch));
-    const size_t shmem = 3 * 2 * WG_TILE_ELEMS * sizeof(unsigned short);
-    hipLaunchKernelGGL(conv_wgrad_kernel, grid, dim3(256), shmem, stream,
-                       (const unsigned short*)dy, (const unsigned short*)x,
-                       ws, N, H, W, C, K, P, Q, R, S, sy, sx, py, px, nch);
+    if (K % 128 == 0 && C % 128 == 0) {
+        dim3 grid((unsigned)(K / 128), (unsigned)(C / 128),
+                  (unsigned)(R * S * nch));
+        const size_t shmem = 2 * 2 * (WG_BM * 128) * sizeof(unsigned short);
+        hipLaunchKernelGGL(conv_wgrad_kernel<4>, grid, dim3(256), shmem,
+                           stream, (const unsigned short*)dy,
+                           (const unsigned short*)x, ws, N, H, W, C, K, P, Q,
+                           R, S, sy, sx, py, px, nch);
+    } else {
+        dim3 grid((unsigned)(K / 64), (unsigned)(C / 64),
+                  (unsigned)(R * S * nch));
+        const size_t shmem = 3 * 2 * (WG_BM * 64) * sizeof(unsigned short);
+        hipLaunchKernelGGL(conv_wgrad_kernel<2>, grid, dim3(256), shmem,
+                           stream, (const unsigned short*)dy,
+                           (const unsigned short*)x, ws, N, H, W, C, K, P, Q,
+                           R, S, sy, sx, py, px, nch);
+    }
 }
 
 }  // namespace fda
