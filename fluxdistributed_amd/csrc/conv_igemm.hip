@@ -564,7 +564,10 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
                        hipStream_t stream) {
     const long M = (long)N * P * Q;
     const int nch = (int)((M + WG_MCH - 1) / WG_MCH);
-    if (K % 128 == 0 && C % 128 == 0) {
+    // FT=4 quarters the block count; only worth it when the grid still
+    // fills the 256 CUs (small-M 1x1 shapes measured 1.4x slower on it)
+    const long blocks4 = (long)(K / 128) * (C / 128) * R * S * nch;
+    if (K % 128 == 0 && C % 128 == 0 && blocks4 >= 192) {
         dim3 grid((unsigned)(K / 128), (unsigned)(C / 128),
                   (unsigned)(R * S * nch));
         const size_t shmem = 2 * 2 * (WG_BM * 128) * sizeof(unsigned short);
