@@ -285,6 +285,27 @@ at::Tensor conv_igemm_wgrad(at::Tensor dy, at::Tensor x,
     return ws;
 }
 
+void conv_igemm_wgrad_into(at::Tensor dy, at::Tensor x, at::Tensor ws,
+                           int64_t R, int64_t S,
+                           int64_t sy, int64_t sx, int64_t py, int64_t px) {
+    // Accumulating variant: ws is a PRE-ZEROED [K, R*S*C] fp32 slice of the
+    // step-scoped wgrad arena (ops/conv.py) — saves the per-layer zeros
+    // launch.
+    TORCH_CHECK(dy.scalar_type() == at::kBFloat16 &&
+                x.scalar_type() == at::kBFloat16);
+    TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                x.is_contiguous(at::MemoryFormat::ChannelsLast));
+    TORCH_CHECK(ws.scalar_type() == at::kFloat && ws.is_contiguous());
+    const int N = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+              W = (int)x.size(3);
+    const int K = (int)dy.size(1), P = (int)dy.size(2), Q = (int)dy.size(3);
+    TORCH_CHECK(C % 64 == 0 && K % 64 == 0);
+    TORCH_CHECK(ws.numel() == (int64_t)K * R * S * C);
+    fda::conv_wgrad_launch(dy.data_ptr(), x.data_ptr(), ws.data_ptr<float>(),
+                           N, H, W, C, K, P, Q, (int)R, (int)S, (int)sy,
+                           (int)sx, (int)py, (int)px, cur_stream());
+}
+
 void wt_transpose_batch(at::Tensor src_ptrs, at::Tensor dst_ptrs,
                         at::Tensor Ks, at::Tensor RCs, at::Tensor tile_counts,
                         int64_t max_tiles) {
@@ -310,6 +331,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_igemm_fwd", &conv_igemm_fwd,
           "implicit-GEMM conv fwd (NHWC bf16, MFMA)");
     m.def("wt_transpose_batch", &wt_transpose_batch);
+    m.def("conv_igemm_wgrad_into", &conv_igemm_wgrad_into);
     m.def("conv_igemm_wgrad", &conv_igemm_wgrad,
           "implicit-GEMM conv weight-grad (NHWC bf16, MFMA + tr16 reads)");
     m.def("conv_igemm_dgrad", &conv_igemm_dgrad,

@@ -100,8 +100,49 @@ class _WtArena:
         return self.arena[off : off + RC * K].view(RC, K)
 
 
+class _WgradArena:
+    """Step-scoped fp32 wgrad workspace: zeroed in ONE launch per backward
+    epoch instead of a per-layer `zeros` each (the wgrad kernel accumulates
+    into its slice with atomicAdd). A second wgrad call for the same weight
+    inside one epoch (gradient micro-accumulation) re-zeros just its slice
+    so the returned dw stays the per-backward gradient."""
+
+    def __init__(self, device):
+        self.device = device
+        self.slices = {}               # id(weight) -> (offset, numel)
+        self.sizes = []
+        self.arena = None
+        self.zero_epoch = None
+        self.seen = set()
+
+    def get(self, weight):
+        K, Cin, R, S = weight.shape
+        n = K * Cin * R * S
+        if id(weight) not in self.slices:
+            self.slices[id(weight)] = (sum(self.sizes), n)
+            self.sizes.append(n)
+            self.arena = None
+        if self.arena is None:
+            total = sum(self.sizes)
+            self.arena = torch.empty(total, dtype=torch.float32,
+                                     device=self.device)
+            self.zero_epoch = None
+        epoch = _WT_MARKER[0]
+        off, n = self.slices[id(weight)]
+        sl = self.arena[off : off + n]
+        if self.zero_epoch != epoch:
+            self.arena.zero_()
+            self.zero_epoch = epoch
+            self.seen = set()
+        elif id(weight) in self.seen:
+            sl.zero_()
+        self.seen.add(id(weight))
+        return sl.view(K, R * S * Cin)
+
+
 _WT_MARKER = [0]
 _ARENAS: dict = {}
+_WS_ARENAS: dict = {}
 
 
 def bump_conv_wt_marker():
@@ -157,7 +198,14 @@ class _FdaConv2d(torch.autograd.Function):
             K, Cin, R, S = w.shape
             if os.environ.get("FLUXDIST_WGRAD", "") != "miopen":
                 C = require_native("conv_igemm_wgrad")
-                ws = C.conv_igemm_wgrad(gy, x, R, S, sy, sx, py, px)
+                if cacheable:
+                    dev = w.device
+                    if dev not in _WS_ARENAS:
+                        _WS_ARENAS[dev] = _WgradArena(dev)
+                    ws = _WS_ARENAS[dev].get(w)
+                    C.conv_igemm_wgrad_into(gy, x, ws, R, S, sy, sx, py, px)
+                else:
+                    ws = C.conv_igemm_wgrad(gy, x, R, S, sy, sx, py, px)
                 # ws [K][R*S*C] fp32 is exactly the channels_last weight
                 # memory order [K][R][S][C]: one flat cast, zero-copy view
                 dw = (ws.to(torch.bfloat16).view(K, R, S, Cin)
