@@ -306,6 +306,40 @@ void conv_igemm_wgrad_into(at::Tensor dy, at::Tensor x, at::Tensor ws,
                            (int)sx, (int)py, (int)px, cur_stream());
 }
 
+at::Tensor conv_stem_fwd(at::Tensor x8, at::Tensor wpad, int64_t R,
+                         int64_t sy, int64_t sx, int64_t P, int64_t Q) {
+    // x8: [N,8,Hp,Wp] channels_last bf16 (spatially pre-padded, channels
+    // 3..7 zero); wpad: [K][R][64] bf16 (s==7 and c>=3 taps zero).
+    TORCH_CHECK(x8.scalar_type() == at::kBFloat16 &&
+                x8.is_contiguous(at::MemoryFormat::ChannelsLast));
+    TORCH_CHECK(x8.size(1) == 8);
+    TORCH_CHECK(wpad.is_contiguous() && wpad.scalar_type() == at::kBFloat16);
+    const int N = (int)x8.size(0), Hp = (int)x8.size(2), Wp = (int)x8.size(3);
+    const int K = (int)wpad.size(0);
+    TORCH_CHECK(K % 64 == 0 && wpad.size(1) == R && wpad.size(2) == 64);
+    auto y = at::empty({N, K, P, Q},
+                       x8.options().memory_format(at::MemoryFormat::ChannelsLast));
+    fda::conv_stem_fwd_launch(x8.data_ptr(), wpad.data_ptr(), y.data_ptr(),
+                              N, Hp, Wp, K, (int)P, (int)Q, (int)R, (int)sy,
+                              (int)sx, cur_stream());
+    return y;
+}
+
+at::Tensor conv_stem_wgrad(at::Tensor dy, at::Tensor x8, int64_t R,
+                           int64_t sy, int64_t sx) {
+    TORCH_CHECK(dy.scalar_type() == at::kBFloat16 &&
+                dy.is_contiguous(at::MemoryFormat::ChannelsLast));
+    TORCH_CHECK(x8.scalar_type() == at::kBFloat16 &&
+                x8.is_contiguous(at::MemoryFormat::ChannelsLast));
+    const int N = (int)x8.size(0), Hp = (int)x8.size(2), Wp = (int)x8.size(3);
+    const int K = (int)dy.size(1), P = (int)dy.size(2), Q = (int)dy.size(3);
+    auto ws = at::zeros({K, R * 64}, x8.options().dtype(at::kFloat));
+    fda::conv_stem_wgrad_launch(dy.data_ptr(), x8.data_ptr(),
+                                ws.data_ptr<float>(), N, Hp, Wp, K, P, Q,
+                                (int)R, (int)sy, (int)sx, cur_stream());
+    return ws;   // [K][R*64]; host slices [K][R][s<7][c<3]
+}
+
 void wt_transpose_batch(at::Tensor src_ptrs, at::Tensor dst_ptrs,
                         at::Tensor Ks, at::Tensor RCs, at::Tensor tile_counts,
                         int64_t max_tiles) {
@@ -331,6 +365,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_igemm_fwd", &conv_igemm_fwd,
           "implicit-GEMM conv fwd (NHWC bf16, MFMA)");
     m.def("wt_transpose_batch", &wt_transpose_batch);
+    m.def("conv_stem_fwd", &conv_stem_fwd);
+    m.def("conv_stem_wgrad", &conv_stem_wgrad);
     m.def("conv_igemm_wgrad_into", &conv_igemm_wgrad_into);
     m.def("conv_igemm_wgrad", &conv_igemm_wgrad,
           "implicit-GEMM conv weight-grad (NHWC bf16, MFMA + tr16 reads)");

@@ -48,7 +48,11 @@ __device__ __align__(16) static const unsigned short conv_zero16[8] = {0};
         (const __attribute__((address_space(1))) void*)(gptr),              \
         (__attribute__((address_space(3))) void*)(lptr), 16, 0, 0)
 
-enum ConvMode { CONV_FWD = 0, CONV_DGRAD = 1 };
+enum ConvMode { CONV_FWD = 0, CONV_DGRAD = 1, CONV_STEM = 2 };
+// CONV_STEM: small-C stem conv on a channel-padded (C=8) pre-padded input.
+// Taps are r-only: each K-step covers one filter row r as 64 virtual
+// channels = 8 pixels (s=0..7, s==7 zero-padded in wpad) x 8 channels
+// (c>=3 zero). B reads wpad[k][r][64]; A granules are whole 16-B pixels.
 
 constexpr int BK = 64;
 
@@ -66,12 +70,13 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     constexpr int AI = BM / 32;          // A glds per wave per tile
     constexpr int BI = BN / 32;          // B glds per wave per tile
 
-    const int OC = (MODE == CONV_FWD) ? K : C;
-    const int RC = (MODE == CONV_FWD) ? C : K;
+    const int OC = (MODE == CONV_DGRAD) ? C : K;
+    const int RC = (MODE == CONV_FWD) ? C : ((MODE == CONV_STEM) ? 64 : K);
 
     int a = 0, b = 0, OH, OW, r0 = 0, s0 = 0, nR = R, nS = S;
-    if (MODE == CONV_FWD) {
+    if (MODE != CONV_DGRAD) {
         OH = P; OW = Q;
+        if (MODE == CONV_STEM) nS = 1;   // taps iterate r only
     } else {
         a = blockIdx.z / sx;  b = blockIdx.z % sx;
         OH = (H - a + sy - 1) / sy;
@@ -112,7 +117,11 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
         const int oh = (int)((mm / OW) % OH);
         const int n = (int)(mm / ((long)OW * OH));
         a_mok[i] = mok;
-        if (MODE == CONV_FWD) {
+        if (MODE == CONV_STEM) {
+            a_hb[i] = oh * sy;           // input pre-padded: no -py
+            a_wb[i] = ow * sx;
+            a_pix[i] = ((long)n * H) * W * 8;
+        } else if (MODE == CONV_FWD) {
             a_hb[i] = oh * sy - py;
             a_wb[i] = ow * sx - px;
             a_pix[i] = ((long)n * H) * W * C;
@@ -141,7 +150,11 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
             const unsigned short* sp;
             bool ok = a_mok[i];
             long off = 0;
-            if (MODE == CONV_FWD) {
+            if (MODE == CONV_STEM) {
+                // 64 virtual channels of K-step ri = pixels (h, w..w+7)x8ch,
+                // all in-bounds on the padded image
+                off = a_pix[i] + ((long)(a_hb[i] + ri) * W + a_wb[i]) * 8 + cs;
+            } else if (MODE == CONV_FWD) {
                 const int h = a_hb[i] + ri, w = a_wb[i] + si;
                 ok = ok && (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W;
                 off = a_pix[i] + ((long)h * W + w) * C + cb + cs;
@@ -161,7 +174,10 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
             const int row = b_row[i];
             const int cs = (cslot ^ (row & 7)) * 8;
             const unsigned short* sp;
-            if (MODE == CONV_FWD) {
+            if (MODE == CONV_STEM) {
+                // wpad[k][r][64]
+                sp = wgt + ((long)(n0 + row) * R + ri) * 64 + cs;
+            } else if (MODE == CONV_FWD) {
                 sp = wgt + ((long)(n0 + row) * R * S * C + (long)rs * C + cb + cs);
             } else {
                 sp = wgt + ((long)((long)rs * C + n0 + row) * K + cb + cs);
@@ -240,7 +256,7 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
             const long m = m0 + wm * 64 + mi * 16 + frow0 + j;
             if (m >= M) continue;
             long obase;
-            if (MODE == CONV_FWD) {
+            if (MODE != CONV_DGRAD) {
                 obase = m * OC;
             } else {
                 const int ww = (int)(m % OW);
@@ -262,17 +278,26 @@ static void launch_cfg(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
                        hipStream_t stream) {
-    const int OC = (MODE == CONV_FWD) ? K : C;
-    const long M = (MODE == CONV_FWD)
+    const int OC = (MODE == CONV_DGRAD) ? C : K;
+    const long M = (MODE != CONV_DGRAD)
         ? (long)N * P * Q
         : (long)N * ((H + sy - 1) / sy) * ((W + sx - 1) / sx);
     dim3 grid((unsigned)((M + BM - 1) / BM), (unsigned)(OC / BN),
-              (MODE == CONV_FWD) ? 1u : (unsigned)(sy * sx));
+              (MODE != CONV_DGRAD) ? 1u : (unsigned)(sy * sx));
     const size_t shmem = 2 * (BM * BK + BN * BK) * sizeof(unsigned short);
     hipLaunchKernelGGL((conv_igemm_kernel<MODE, BM, BN, WN>), grid, dim3(256),
                        shmem, stream, (const unsigned short*)src,
                        (const unsigned short*)wgt, (unsigned short*)out,
                        N, H, W, C, K, P, Q, R, S, sy, sx, py, px);
+}
+
+void conv_stem_fwd_launch(const void* src, const void* wgt, void* out,
+                          int N, int Hp, int Wp, int K, int P, int Q,
+                          int R, int sy, int sx, hipStream_t stream) {
+    // src: channel-padded (C=8) spatially pre-padded input [N,Hp,Wp,8];
+    // wgt: wpad [K][R][64]; out: [N*P*Q][K]
+    launch_cfg<CONV_STEM, 256, 64, 1>(src, wgt, out, N, Hp, Wp, /*C=*/8, K,
+                                      P, Q, R, /*S=*/1, sy, sx, 0, 0, stream);
 }
 
 void conv_igemm_launch(const void* src, const void* wgt, void* out,
@@ -381,7 +406,7 @@ typedef __attribute__((ext_vector_type(4))) short short4_;
 constexpr int WG_BM = 64;      // m per K-step
 constexpr int WG_MCH = 2048;   // pixels per block (chunk)
 
-template <int FT>
+template <int FT, bool STEM = false>
 __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
     const unsigned short* __restrict__ dy,   // [M][K] (NHWC out grad)
     const unsigned short* __restrict__ x,    // [N,H,W,C]
@@ -452,11 +477,19 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
         for (int i = 0; i < FT; ++i) {
             const unsigned short* sp = conv_zero16;
             if (st_m[i] < mend) {
-                const int hh = st_p[i] * sy - py + r;
-                const int ww = st_q[i] * sx - px + s;
-                if ((unsigned)hh < (unsigned)H && (unsigned)ww < (unsigned)W)
-                    sp = x + (((long)st_n[i] * H + hh) * W + ww) * C +
-                         c0 + st_ch[i];
+                if (STEM) {
+                    // pre-padded C=8 image; virtual channel = pixel s x 8ch
+                    const int hh = st_p[i] * sy + r;
+                    const int ww = st_q[i] * sx;
+                    sp = x + (((long)st_n[i] * H + hh) * W + ww) * 8 +
+                         st_ch[i];
+                } else {
+                    const int hh = st_p[i] * sy - py + r;
+                    const int ww = st_q[i] * sx - px + s;
+                    if ((unsigned)hh < (unsigned)H && (unsigned)ww < (unsigned)W)
+                        sp = x + (((long)st_n[i] * H + hh) * W + ww) * C +
+                             c0 + st_ch[i];
+                }
             }
             FDA_GLDS16(sp, base + TILE_ELEMS + (wid * FT + i) * 512);
         }
@@ -556,6 +589,20 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
                 atomicAdd(&ws[kk * RSC + (long)rs * C + cc],
                           acc[ki][ci][jj]);
             }
+}
+
+void conv_stem_wgrad_launch(const void* dy, const void* x, float* ws,
+                            int N, int Hp, int Wp, int K, int P, int Q,
+                            int R, int sy, int sx, hipStream_t stream) {
+    // dy [M][K]; x padded C=8 image; ws [K][R*64] fp32 pre-zeroed.
+    const long M = (long)N * P * Q;
+    const int nch = (int)((M + WG_MCH - 1) / WG_MCH);
+    dim3 grid((unsigned)(K / 64), 1u, (unsigned)(R * nch));
+    const size_t shmem = 3 * 2 * (WG_BM * 64) * sizeof(unsigned short);
+    hipLaunchKernelGGL((conv_wgrad_kernel<2, true>), grid, dim3(256), shmem,
+                       stream, (const unsigned short*)dy,
+                       (const unsigned short*)x, ws, N, Hp, Wp, /*C=*/64, K,
+                       P, Q, R, /*S=*/1, sy, sx, 0, 0, nch);
 }
 
 void conv_wgrad_launch(const void* dy, const void* x, float* ws,

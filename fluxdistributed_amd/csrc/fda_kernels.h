@@ -71,6 +71,14 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int R, int S, int sy, int sx, int py, int px,
                        bool dgrad, hipStream_t stream);
 
+// stem conv (small C via channel-pad to 8, spatially pre-padded input)
+void conv_stem_fwd_launch(const void* src, const void* wgt, void* out,
+                          int N, int Hp, int Wp, int K, int P, int Q,
+                          int R, int sy, int sx, hipStream_t stream);
+void conv_stem_wgrad_launch(const void* dy, const void* x, float* ws,
+                            int N, int Hp, int Wp, int K, int P, int Q,
+                            int R, int sy, int sx, hipStream_t stream);
+
 // wgrad: ws[K][RS*C] fp32 (pre-zeroed) += dy^T @ im2col(x), atomic chunks
 void conv_wgrad_launch(const void* dy, const void* x, float* ws,
                        int N, int H, int W, int C, int K, int P, int Q,

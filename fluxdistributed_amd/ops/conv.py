@@ -217,12 +217,71 @@ class _FdaConv2d(torch.autograd.Function):
         return dx, dw, None, None
 
 
+class _FdaStemConv2d(torch.autograd.Function):
+    """Small-C stem conv (C<=5, S<=7): channel-pad to 8 + spatial pre-pad,
+    then the CONV_STEM kernel (one K-step per filter row r — 8 pixels x 8
+    channels as 64 virtual reduction channels). No dgrad (the stem input is
+    data); wgrad via conv_stem_wgrad."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, padding):
+        C = require_native("conv_stem_fwd")
+        K, Cin, R, S = weight.shape
+        sy, sx = stride
+        py, px = padding
+        N, _, H, W = x.shape
+        P = (H + 2 * py - R) // sy + 1
+        Q = (W + 2 * px - S) // sx + 1
+        xc = x.contiguous(memory_format=torch.channels_last)
+        # pad channels to 8 and spatial by (py, px); extra right-edge pixel
+        # slack so the 8-pixel (s=0..7) granule row never leaves the image
+        Wp = W + 2 * px + 8
+        x8 = torch.zeros(N, 8, H + 2 * py, Wp, dtype=x.dtype, device=x.device)
+        x8 = x8.contiguous(memory_format=torch.channels_last)
+        x8[:, :Cin, py : py + H, px : px + W] = xc
+        wpad = torch.zeros(K, R, 64, dtype=weight.dtype, device=weight.device)
+        wpad.view(K, R, 8, 8)[:, :, :S, :Cin] = (
+            weight.contiguous(memory_format=torch.channels_last)
+            .permute(0, 2, 3, 1))  # [K][R][S][C]
+        y = C.conv_stem_fwd(x8, wpad, R, sy, sx, P, Q)
+        ctx.save_for_backward(x8)
+        ctx.conf = (K, Cin, R, S, sy, sx)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        (x8,) = ctx.saved_tensors
+        K, Cin, R, S, sy, sx = ctx.conf
+        dw = None
+        if ctx.needs_input_grad[1]:
+            C = require_native("conv_stem_wgrad")
+            gyc = gy.contiguous(memory_format=torch.channels_last)
+            ws = C.conv_stem_wgrad(gyc, x8, R, sy, sx)
+            # ws [K][R*64] -> [K][R][s][c] -> weight grad [K,C,R,S]
+            dw = (ws.view(K, R, 8, 8)[:, :, :S, :Cin]
+                  .to(torch.bfloat16).permute(0, 3, 1, 2))
+        return None, dw, None, None
+
+
+def _stem_supported(x, weight, stride, padding, dilation, groups) -> bool:
+    if os.environ.get("FLUXDIST_CONV", "") == "miopen":
+        return False
+    if not (x.is_cuda and x.dtype == torch.bfloat16):
+        return False
+    if groups != 1 or _pair(dilation) != (1, 1):
+        return False
+    K, Cin, R, S = weight.shape
+    return Cin <= 5 and S <= 7 and K % 64 == 0 and load_native() is not None
+
+
 def fda_conv2d(x: torch.Tensor, weight: torch.Tensor, stride=(1, 1),
                padding=(0, 0), dilation=(1, 1), groups: int = 1) -> torch.Tensor:
     """conv2d with per-shape dispatch to the native MFMA kernel."""
     stride, padding, dilation = _pair(stride), _pair(padding), _pair(dilation)
     if _native_supported(x, weight, stride, padding, dilation, groups):
         return _FdaConv2d.apply(x, weight, stride, padding)
+    if _stem_supported(x, weight, stride, padding, dilation, groups):
+        return _FdaStemConv2d.apply(x, weight, stride, padding)
     if os.environ.get("FLUXDIST_CONV", "") == "fda" and x.is_cuda:
         raise RuntimeError(
             f"FLUXDIST_CONV=fda but shape unsupported by conv_igemm: "

@@ -125,3 +125,46 @@ def test_conv_wgrad_matches_fp32(shape):
     scale = dw_ref.abs().max().item()
     # reduction over N*P*Q in bf16 products, fp32 accumulate
     assert err < 0.02 * max(scale, 1.0), f"{shape}: err={err} scale={scale}"
+
+
+STEM_SHAPES = [
+    (4, 3, 224, 224, 64, 7, 2, 3),   # ImageNet stem
+    (8, 3, 32, 32, 64, 3, 1, 1),     # CIFAR stem
+]
+
+
+@pytest.mark.parametrize("shape", STEM_SHAPES)
+def test_stem_conv_fwd_and_wgrad(shape):
+    from fluxdistributed_amd.ops.conv import _FdaStemConv2d, _stem_supported
+
+    n, c, h, w, k, r, s, pad = shape
+    x, wt = _mk(n, c, h, w, k, r, seed=5)
+    w32 = wt.clone().requires_grad_(True)
+    ref = torch.nn.functional.conv2d(x, w32, stride=s, padding=pad)
+    gy = torch.randn_like(ref)
+    ref.backward(gy)
+
+    xb = x.bfloat16().contiguous(memory_format=torch.channels_last)
+    wb = wt.bfloat16().contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    assert _stem_supported(xb, wb, (s, s), (pad, pad), (1, 1), 1)
+    y = _FdaStemConv2d.apply(xb, wb, (s, s), (pad, pad))
+    err = (y.float() - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 0.03 * max(scale, 1.0), f"fwd err={err} scale={scale}"
+
+    y.backward(gy.bfloat16().contiguous(memory_format=torch.channels_last))
+    errw = (wb.grad.float() - w32.grad).abs().max().item()
+    scalew = w32.grad.abs().max().item()
+    assert errw < 0.02 * max(scalew, 1.0), f"wgrad err={errw} scale={scalew}"
+
+
+def test_stem_dispatch_in_model_forward():
+    from fluxdistributed_amd.ops.conv import fda_conv2d
+
+    x = torch.randn(2, 3, 64, 64).cuda().bfloat16() \
+        .contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(64, 3, 7, 7) * 0.05).cuda().bfloat16() \
+        .contiguous(memory_format=torch.channels_last)
+    y = fda_conv2d(x, w, (2, 2), (3, 3))
+    ref = torch.nn.functional.conv2d(x.float(), w.float(), stride=2, padding=3)
+    assert (y.float() - ref).abs().max().item() < 0.05 * ref.abs().max().item()
