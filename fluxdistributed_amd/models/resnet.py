@@ -135,11 +135,13 @@ class ResNet(nn.Module):
         super().__init__()
         self.small_input = small_input
         self.cin = 64
+        from ..ops.conv import FdaConv2d
+
         if small_input:  # CIFAR-style stem (BASELINE config 1)
-            self.conv1 = nn.Conv2d(3, 64, 3, stride=1, padding=1, bias=False)
+            self.conv1 = FdaConv2d(3, 64, 3, stride=1, padding=1, bias=False)
             self.maxpool = nn.Identity()
         else:
-            self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+            self.conv1 = FdaConv2d(3, 64, 7, stride=2, padding=3, bias=False)
             self.maxpool = MaxPool2d(3, stride=2, padding=1)
         self.bn1 = FusedBNAct(64, relu=True)
         self.layer1 = self._make_layer(block, 64, layers[0])
