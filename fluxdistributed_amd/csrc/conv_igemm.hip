@@ -412,7 +412,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
     const unsigned short* __restrict__ x,    // [N,H,W,C]
     float* __restrict__ ws,                  // [K][RS*C] fp32, pre-zeroed
     int N, int H, int W, int C, int K, int P, int Q,
-    int R, int S, int sy, int sx, int py, int px, int nch) {
+    int R, int S, int sy, int sx, int py, int px, int nch, int mch) {
     constexpr int TCH = 32 * FT;             // tile channels per operand
     constexpr int TILE_ELEMS = WG_BM * TCH;  // one operand tile
     const int rs = blockIdx.z / nch;
@@ -421,8 +421,8 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
     const int k0 = blockIdx.x * TCH;
     const int c0 = blockIdx.y * TCH;
     const long M = (long)N * P * Q;
-    const long mb0 = (long)chunk * WG_MCH;
-    const long mend = (mb0 + WG_MCH < M) ? mb0 + WG_MCH : M;
+    const long mb0 = (long)chunk * mch;
+    const long mend = (mb0 + mch < M) ? mb0 + mch : M;
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -595,14 +595,18 @@ void conv_stem_wgrad_launch(const void* dy, const void* x, float* ws,
                             int N, int Hp, int Wp, int K, int P, int Q,
                             int R, int sy, int sx, hipStream_t stream) {
     // dy [M][K]; x padded C=8 image; ws [K][R*64] fp32 pre-zeroed.
+    // Output is tiny (K x R*64), so small chunks would hammer the same
+    // fp32 addresses with atomics (588 hits/address measured at 2048):
+    // use 16k-pixel chunks.
     const long M = (long)N * P * Q;
-    const int nch = (int)((M + WG_MCH - 1) / WG_MCH);
+    constexpr int STEM_MCH = 8 * WG_MCH;
+    const int nch = (int)((M + STEM_MCH - 1) / STEM_MCH);
     dim3 grid((unsigned)(K / 64), 1u, (unsigned)(R * nch));
     const size_t shmem = 3 * 2 * (WG_BM * 64) * sizeof(unsigned short);
     hipLaunchKernelGGL((conv_wgrad_kernel<2, true>), grid, dim3(256), shmem,
                        stream, (const unsigned short*)dy,
                        (const unsigned short*)x, ws, N, Hp, Wp, /*C=*/64, K,
-                       P, Q, R, /*S=*/1, sy, sx, 0, 0, nch);
+                       P, Q, R, /*S=*/1, sy, sx, 0, 0, nch, STEM_MCH);
 }
 
 void conv_wgrad_launch(const void* dy, const void* x, float* ws,
@@ -621,7 +625,7 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
         hipLaunchKernelGGL(conv_wgrad_kernel<4>, grid, dim3(256), shmem,
                            stream, (const unsigned short*)dy,
                            (const unsigned short*)x, ws, N, H, W, C, K, P, Q,
-                           R, S, sy, sx, py, px, nch);
+                           R, S, sy, sx, py, px, nch, WG_MCH);
     } else {
         dim3 grid((unsigned)(K / 64), (unsigned)(C / 64),
                   (unsigned)(R * S * nch));
@@ -629,7 +633,7 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
         hipLaunchKernelGGL(conv_wgrad_kernel<2>, grid, dim3(256), shmem,
                            stream, (const unsigned short*)dy,
                            (const unsigned short*)x, ws, N, H, W, C, K, P, Q,
-                           R, S, sy, sx, py, px, nch);
+                           R, S, sy, sx, py, px, nch, WG_MCH);
     }
 }
 

@@ -143,6 +143,7 @@ class _WgradArena:
 _WT_MARKER = [0]
 _ARENAS: dict = {}
 _WS_ARENAS: dict = {}
+_STEM_X8: dict = {}
 
 
 def bump_conv_wt_marker():
@@ -234,23 +235,33 @@ class _FdaStemConv2d(torch.autograd.Function):
         Q = (W + 2 * px - S) // sx + 1
         xc = x.contiguous(memory_format=torch.channels_last)
         # pad channels to 8 and spatial by (py, px); extra right-edge pixel
-        # slack so the 8-pixel (s=0..7) granule row never leaves the image
+        # slack so the 8-pixel (s=0..7) granule row never leaves the image.
+        # The padded buffer is cached per shape: the zero border never
+        # changes, so each step only copies the interior.
         Wp = W + 2 * px + 8
-        x8 = torch.zeros(N, 8, H + 2 * py, Wp, dtype=x.dtype, device=x.device)
-        x8 = x8.contiguous(memory_format=torch.channels_last)
+        key = (N, Cin, H, W, py, px, x.device)
+        x8 = _STEM_X8.get(key)
+        if x8 is None:
+            x8 = torch.empty(N, 8, H + 2 * py, Wp, dtype=x.dtype,
+                             device=x.device,
+                             memory_format=torch.channels_last).zero_()
+            _STEM_X8[key] = x8
         x8[:, :Cin, py : py + H, px : px + W] = xc
         wpad = torch.zeros(K, R, 64, dtype=weight.dtype, device=weight.device)
         wpad.view(K, R, 8, 8)[:, :, :S, :Cin] = (
             weight.contiguous(memory_format=torch.channels_last)
             .permute(0, 2, 3, 1))  # [K][R][S][C]
         y = C.conv_stem_fwd(x8, wpad, R, sy, sx, P, Q)
-        ctx.save_for_backward(x8)
+        # plain attribute, not save_for_backward: x8 is a reused cached
+        # buffer whose interior is rewritten every forward; the version-
+        # counter check would reject the standard fwd->bwd->fwd loop.
+        ctx.x8 = x8
         ctx.conf = (K, Cin, R, S, sy, sx)
         return y
 
     @staticmethod
     def backward(ctx, gy):
-        (x8,) = ctx.saved_tensors
+        x8 = ctx.x8
         K, Cin, R, S, sy, sx = ctx.conf
         dw = None
         if ctx.needs_input_grad[1]:
