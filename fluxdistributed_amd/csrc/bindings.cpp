@@ -121,13 +121,16 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
     at::Tensor gout, at::Tensor x, at::Tensor weight, at::Tensor save_mean,
-    at::Tensor save_invstd, at::Tensor out, bool relu, bool training) {
+    at::Tensor save_invstd, at::Tensor out, bool relu, bool training,
+    at::Tensor gw_out, at::Tensor gb_out) {
+    // gw_out/gb_out non-empty: fp32 flat-G slices, written += (direct grad)
     auto [rows, C] = nhwc_rows(x);
     auto gc = gout.contiguous(at::MemoryFormat::ChannelsLast);
     auto fopts = x.options().dtype(at::kFloat);
     auto ws = at::empty({4 * C}, fopts);
-    auto gw = at::empty({C}, fopts);
-    auto gb = at::empty({C}, fopts);
+    const bool direct = gw_out.defined() && gw_out.numel() == C;
+    auto gw = direct ? gw_out : at::empty({C}, fopts);
+    auto gb = direct ? gb_out : at::empty({C}, fopts);
     auto gx = at::empty_like(x);
     auto stream = cur_stream();
 
@@ -139,7 +142,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
                              save_invstd.data_ptr<float>(), ws.data_ptr<float>(),
                              part.data_ptr<float>(), gw.data_ptr<float>(),
                              gb.data_ptr<float>(), rows, (int)C, relu,
-                             training, dt_of(x), stream);
+                             training, direct, dt_of(x), stream);
     fda::bn_bwd_apply_launch(gc.data_ptr(), x.data_ptr(), out.data_ptr(),
                              save_mean.data_ptr<float>(),
                              save_invstd.data_ptr<float>(),
@@ -340,6 +343,14 @@ at::Tensor conv_stem_wgrad(at::Tensor dy, at::Tensor x8, int64_t R,
     return ws;   // [K][R*64]; host slices [K][R][s<7][c<3]
 }
 
+void grad_accum_bf16(at::Tensor g, at::Tensor ws) {
+    TORCH_CHECK(g.scalar_type() == at::kBFloat16 && g.is_contiguous());
+    TORCH_CHECK(ws.scalar_type() == at::kFloat && ws.is_contiguous());
+    TORCH_CHECK(g.numel() == ws.numel());
+    fda::grad_accum_bf16_launch(g.data_ptr(), ws.data_ptr<float>(),
+                                (long)g.numel(), cur_stream());
+}
+
 void wt_transpose_batch(at::Tensor src_ptrs, at::Tensor dst_ptrs,
                         at::Tensor Ks, at::Tensor RCs, at::Tensor tile_counts,
                         int64_t max_tiles) {
@@ -365,6 +376,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_igemm_fwd", &conv_igemm_fwd,
           "implicit-GEMM conv fwd (NHWC bf16, MFMA)");
     m.def("wt_transpose_batch", &wt_transpose_batch);
+    m.def("grad_accum_bf16", &grad_accum_bf16);
     m.def("conv_stem_fwd", &conv_stem_fwd);
     m.def("conv_stem_wgrad", &conv_stem_wgrad);
     m.def("conv_igemm_wgrad_into", &conv_igemm_wgrad_into);

@@ -154,10 +154,11 @@ __global__ void bn_fwd_reduce_finalize_kernel(
 __global__ void bn_bwd_reduce_finalize_kernel(
     const float* __restrict__ part, float* __restrict__ ws,
     float* __restrict__ gw, float* __restrict__ gb, int NB, int chunkC,
-    int c_base, int C, int64_t rows, int training) {
+    int c_base, int C, int64_t rows, int training, int accum) {
     bn_reduce_then(part, NB, chunkC, c_base, [&](int c, BnTotals t) {
-        gb[c] = t.s;                       // sum_g
-        gw[c] = t.q;                       // sum_g_xhat
+        // accum: gw/gb are flat-G slices (direct grad, += semantics)
+        gb[c] = accum ? gb[c] + t.s : t.s;  // sum_g
+        gw[c] = accum ? gw[c] + t.q : t.q;  // sum_g_xhat
         const float inv_m = training ? 1.f / (float)rows : 0.f;
         ws[2 * C + c] = t.s * inv_m;       // k1
         ws[3 * C + c] = t.q * inv_m;       // k2
@@ -388,8 +389,8 @@ void bn_apply_launch(const void* x, const void* residual, void* out,
 void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
                          const float* save_mean, const float* save_invstd,
                          float* ws, float* part, float* gw, float* gb,
-                         int64_t rows, int C, bool relu, bool training, DT dt,
-                         hipStream_t s) {
+                         int64_t rows, int C, bool relu, bool training,
+                         bool accum_g, DT dt, hipStream_t s) {
     const int V = dt == DT::BF16 ? 8 : 4;
     int chunkC, nchunks, grid, shmem;
     stats_geom(C, V, rows, chunkC, nchunks, grid, shmem);
@@ -406,7 +407,8 @@ void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
         #undef FDA_BSTATS
         hipLaunchKernelGGL(bn_bwd_reduce_finalize_kernel, dim3(cc / 64),
                            dim3(256), 0, s, part, ws, gw, gb, grid, cc,
-                           c_base, C, rows, training ? 1 : 0);
+                           c_base, C, rows, training ? 1 : 0,
+                           accum_g ? 1 : 0);
     }
 }
 

@@ -276,3 +276,35 @@ void adam_step_launch(void* P, const void* G, float* M, float* V, float* S,
 }
 
 }  // namespace fda
+
+namespace fda {
+
+// G_bf16 += cast(ws_f32): the direct-grad flush for one conv weight slice
+// (replaces an aten cast kernel + an AccumulateGrad add per layer).
+__global__ __launch_bounds__(256) void grad_accum_bf16_kernel(
+    unsigned short* __restrict__ g, const float* __restrict__ ws, long n) {
+    const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (i0 + 8 <= n) {
+        unsigned short gv[8];
+        *(uint4*)gv = *(const uint4*)(g + i0);
+        float wv[8];
+        *(float4*)(wv) = *(const float4*)(ws + i0);
+        *(float4*)(wv + 4) = *(const float4*)(ws + i0 + 4);
+        #pragma unroll
+        for (int e = 0; e < 8; ++e)
+            gv[e] = f32_to_bf16bits(bf16bits_to_f32(gv[e]) + wv[e]);
+        *(uint4*)(g + i0) = *(const uint4*)gv;
+    } else {
+        for (long i = i0; i < n; ++i)
+            g[i] = f32_to_bf16bits(bf16bits_to_f32(g[i]) + ws[i]);
+    }
+}
+
+void grad_accum_bf16_launch(void* g, const float* ws, long n, hipStream_t s) {
+    const long lanes = (n + 7) / 8;
+    dim3 grid((unsigned)((lanes + 255) / 256));
+    hipLaunchKernelGGL(grad_accum_bf16_kernel, grid, dim3(256), 0, s,
+                       (unsigned short*)g, ws, n);
+}
+
+}  // namespace fda

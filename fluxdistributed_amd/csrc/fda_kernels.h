@@ -47,8 +47,8 @@ void bn_apply_launch(const void* x, const void* residual, void* out,
 void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
                          const float* save_mean, const float* save_invstd,
                          float* ws, float* part, float* gw, float* gb,
-                         int64_t rows, int C, bool relu, bool training, DT dt,
-                         hipStream_t s);
+                         int64_t rows, int C, bool relu, bool training,
+                         bool accum_g, DT dt, hipStream_t s);
 void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
                          const float* save_mean, const float* save_invstd,
                          const float* weight, const float* ws, void* gx,
@@ -90,6 +90,9 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
 void wt_transpose_launch(const long* src_ptrs, long* dst_ptrs, const int* Ks,
                          const int* RCs, const int* tile_counts, int ntensors,
                          int max_tiles, hipStream_t stream);
+
+// direct-grad flush: G_bf16 += cast(ws_f32), one conv-weight slice
+void grad_accum_bf16_launch(void* g, const float* ws, long n, hipStream_t s);
 
 // flat fused optimizers. P: param dtype; M/V/S fp32; G param dtype.
 void sgd_step_launch(void* P, const void* G, float* M, float* V, int64_t n,

@@ -198,7 +198,10 @@ class _FdaConv2d(torch.autograd.Function):
         if ctx.needs_input_grad[1]:
             K, Cin, R, S = w.shape
             if os.environ.get("FLUXDIST_WGRAD", "") != "miopen":
+                from .fused_optim import flat_grad_slice
+
                 C = require_native("conv_igemm_wgrad")
+                g_sl = flat_grad_slice(w) if cacheable else None
                 if cacheable:
                     dev = w.device
                     if dev not in _WS_ARENAS:
@@ -207,10 +210,19 @@ class _FdaConv2d(torch.autograd.Function):
                     C.conv_igemm_wgrad_into(gy, x, ws, R, S, sy, sx, py, px)
                 else:
                     ws = C.conv_igemm_wgrad(gy, x, R, S, sy, sx, py, px)
-                # ws [K][R*S*C] fp32 is exactly the channels_last weight
-                # memory order [K][R][S][C]: one flat cast, zero-copy view
-                dw = (ws.to(torch.bfloat16).view(K, R, S, Cin)
-                      .permute(0, 3, 1, 2))
+                if g_sl is not None:
+                    # direct grad: G slice memory order == ws order
+                    # ([K][R][S][C]); one fused cast+add, no AccumulateGrad
+                    C.grad_accum_bf16(g_sl, ws.reshape(-1))
+                    from ..parallel.bucketing import notify_grad_written
+
+                    notify_grad_written(w)
+                    dw = None
+                else:
+                    # ws [K][R*S*C] fp32 is exactly the channels_last weight
+                    # memory order [K][R][S][C]: one flat cast, no-copy view
+                    dw = (ws.to(torch.bfloat16).view(K, R, S, Cin)
+                          .permute(0, 3, 1, 2))
             else:
                 dw = torch.ops.aten.convolution_backward(
                     gy, x, w, None, list(stride), list(padding), [1, 1],

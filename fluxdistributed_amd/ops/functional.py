@@ -159,6 +159,7 @@ class _BNAct(torch.autograd.Function):
             residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype),
         )
         ctx.save_for_backward(x, weight, save_mean, save_invstd, out)
+        ctx.bn_bias = bias
         ctx.relu = relu
         ctx.has_residual = residual is not None
         ctx.training = training
@@ -167,11 +168,28 @@ class _BNAct(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_out):
         x, weight, save_mean, save_invstd, out = ctx.saved_tensors
+        bias = ctx.bn_bias
         C = require_native("batch_norm_act")
+        # direct grads: when the params live in a fused-optimizer flat
+        # buffer, the finalize kernel += 's into the G slices and no
+        # AccumulateGrad kernels run (ops/fused_optim.FLAT_SLICES).
+        from .fused_optim import flat_grad_slice
+
+        gw_sl = flat_grad_slice(weight)
+        gb_sl = flat_grad_slice(bias) if bias is not None else None
+        direct = gw_sl is not None and gb_sl is not None
+        empty = torch.empty(0, device=x.device, dtype=torch.float32)
         gx, gw, gb = C.bn_act_bwd(
             grad_out.contiguous(memory_format=torch.channels_last),
             x, weight, save_mean, save_invstd, out, ctx.relu, ctx.training,
+            gw_sl if direct else empty, gb_sl if direct else empty,
         )
+        if direct:
+            from ..parallel.bucketing import notify_grad_written
+
+            notify_grad_written(weight)
+            notify_grad_written(bias)
+            gw = gb = None
         gres = gx_res = None
         if ctx.has_residual:
             # d(out)/d(residual) = relu-mask * grad_out = first stage of gx;

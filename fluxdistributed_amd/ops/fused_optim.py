@@ -31,6 +31,21 @@ def _bump_wt_marker():
     bump_conv_wt_marker()
 
 
+# id(param) -> (flat G tensor, offset, numel): lets backward ops write
+# gradients straight into the flat buffer (no AccumulateGrad kernel) —
+# see ops/conv.py and ops/functional.py "direct grad" paths.
+FLAT_SLICES = {}
+
+
+def flat_grad_slice(param):
+    """Contiguous flat-G slice backing param.grad, or None."""
+    ent = FLAT_SLICES.get(id(param))
+    if ent is None:
+        return None
+    G, off, n = ent
+    return G[off : off + n]
+
+
 class _FlatGroup:
     """All parameters of one (device, dtype) flattened into shared storage."""
 
@@ -54,6 +69,7 @@ class _FlatGroup:
             v.copy_(p.data)
             p.data = v
             p.grad = self._view_like(self.G, off, p.data)
+            FLAT_SLICES[id(p)] = (self.G, off, p.numel())
         self.master = self.P.float() if dt != torch.float32 else None
 
     @staticmethod

@@ -21,6 +21,17 @@ import torch.distributed as dist
 
 from ..ops.fused_optim import _FlatOptimizer
 
+# Direct-grad notification: backward ops that write gradients straight into
+# the flat G buffer (bypassing AccumulateGrad, so post-accumulate hooks
+# never fire) call notify_grad_written(param) instead. An attached
+# GradBucketer registers itself here.
+_NOTIFY = []
+
+
+def notify_grad_written(param) -> None:
+    for cb in _NOTIFY:
+        cb(param)
+
 
 class GradBucketer:
     """Slices a flat gradient buffer into buckets keyed to parameters and
@@ -80,17 +91,25 @@ class GradBucketer:
         self._ready = [False] * len(self.buckets)
 
     def attach(self):
-        """Install post-accumulate-grad hooks on every bucketed parameter."""
+        """Install post-accumulate-grad hooks on every bucketed parameter
+        and subscribe to direct-grad notifications."""
         self._handles = []
         for b in self.buckets:
             for p in b["params"]:
                 h = p.register_post_accumulate_grad_hook(self._on_grad)
                 self._handles.append(h)
+        _NOTIFY.append(self._on_direct)
 
     def detach(self):
         for h in getattr(self, "_handles", []):
             h.remove()
         self._handles = []
+        if self._on_direct in _NOTIFY:
+            _NOTIFY.remove(self._on_direct)
+
+    def _on_direct(self, p):
+        if id(p) in self._param2bucket:
+            self._on_grad(p)
 
     def _on_grad(self, p):
         bi = self._param2bucket[id(p)]
