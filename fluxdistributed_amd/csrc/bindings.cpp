@@ -368,7 +368,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("add_relu_fwd", &add_relu_fwd);
     m.def("add_relu_bwd", &add_relu_bwd);
     m.def("bn_act_fwd", &bn_act_fwd);
-    m.def("bn_act_bwd", &bn_act_bwd);
+    m.def("bn_act_bwd", &bn_act_bwd, pybind11::arg("gout"),
+          pybind11::arg("x"), pybind11::arg("weight"),
+          pybind11::arg("save_mean"), pybind11::arg("save_invstd"),
+          pybind11::arg("out"), pybind11::arg("relu"),
+          pybind11::arg("training"),
+          pybind11::arg("gw_out") = at::Tensor(),
+          pybind11::arg("gb_out") = at::Tensor());
     m.def("maxpool_fwd", &maxpool_fwd);
     m.def("maxpool_bwd", &maxpool_bwd);
     m.def("sgd_step", &sgd_step);
