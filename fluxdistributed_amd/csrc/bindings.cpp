@@ -122,15 +122,15 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
 std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
     at::Tensor gout, at::Tensor x, at::Tensor weight, at::Tensor save_mean,
     at::Tensor save_invstd, at::Tensor out, bool relu, bool training,
-    at::Tensor gw_out, at::Tensor gb_out) {
-    // gw_out/gb_out non-empty: fp32 flat-G slices, written += (direct grad)
+    c10::optional<at::Tensor> gw_out, c10::optional<at::Tensor> gb_out) {
+    // gw_out/gb_out set: fp32 flat-G slices, written += (direct grad)
     auto [rows, C] = nhwc_rows(x);
     auto gc = gout.contiguous(at::MemoryFormat::ChannelsLast);
     auto fopts = x.options().dtype(at::kFloat);
     auto ws = at::empty({4 * C}, fopts);
-    const bool direct = gw_out.defined() && gw_out.numel() == C;
-    auto gw = direct ? gw_out : at::empty({C}, fopts);
-    auto gb = direct ? gb_out : at::empty({C}, fopts);
+    const bool direct = gw_out.has_value() && gw_out->numel() == C;
+    auto gw = direct ? *gw_out : at::empty({C}, fopts);
+    auto gb = direct ? *gb_out : at::empty({C}, fopts);
     auto gx = at::empty_like(x);
     auto stream = cur_stream();
 
@@ -373,8 +373,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("save_mean"), pybind11::arg("save_invstd"),
           pybind11::arg("out"), pybind11::arg("relu"),
           pybind11::arg("training"),
-          pybind11::arg("gw_out") = at::Tensor(),
-          pybind11::arg("gb_out") = at::Tensor());
+          pybind11::arg("gw_out") = pybind11::none(),
+          pybind11::arg("gb_out") = pybind11::none());
     m.def("maxpool_fwd", &maxpool_fwd);
     m.def("maxpool_bwd", &maxpool_bwd);
     m.def("sgd_step", &sgd_step);

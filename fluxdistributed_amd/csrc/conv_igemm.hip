@@ -609,31 +609,48 @@ void conv_stem_wgrad_launch(const void* dy, const void* x, float* ws,
                        P, Q, R, /*S=*/1, sy, sx, 0, 0, nch, STEM_MCH);
 }
 
+// pick the pixel-chunk size so the grid lands near `target` blocks:
+// enough parallelism to fill the chip, few enough chunks that the fp32
+// atomic fan-in per output address stays small.
+static int pick_mch(long M, long tiles, int target) {
+    long nch_t = target / (tiles > 0 ? tiles : 1);
+    if (nch_t < 1) nch_t = 1;
+    long mch = (M + nch_t - 1) / nch_t;
+    mch = ((mch + WG_MCH - 1) / WG_MCH) * WG_MCH;   // multiple of 2048
+    if (mch < WG_MCH) mch = WG_MCH;
+    return (int)mch;
+}
+
 void conv_wgrad_launch(const void* dy, const void* x, float* ws,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
                        hipStream_t stream) {
     const long M = (long)N * P * Q;
-    const int nch = (int)((M + WG_MCH - 1) / WG_MCH);
     // FT=4 quarters the block count; only worth it when the grid still
     // fills the 256 CUs (small-M 1x1 shapes measured 1.4x slower on it)
-    const long blocks4 = (long)(K / 128) * (C / 128) * R * S * nch;
-    if (K % 128 == 0 && C % 128 == 0 && blocks4 >= 192) {
+    const long tiles4 = (long)(K / 128) * (C / 128) * R * S;
+    const long blocks4_max = tiles4 * ((M + WG_MCH - 1) / WG_MCH);
+    if (K % 128 == 0 && C % 128 == 0 && blocks4_max >= 192) {
+        const int mch = pick_mch(M, tiles4, 768);
+        const int nch = (int)((M + mch - 1) / mch);
         dim3 grid((unsigned)(K / 128), (unsigned)(C / 128),
                   (unsigned)(R * S * nch));
         const size_t shmem = 2 * 2 * (WG_BM * 128) * sizeof(unsigned short);
         hipLaunchKernelGGL(conv_wgrad_kernel<4>, grid, dim3(256), shmem,
                            stream, (const unsigned short*)dy,
                            (const unsigned short*)x, ws, N, H, W, C, K, P, Q,
-                           R, S, sy, sx, py, px, nch, WG_MCH);
+                           R, S, sy, sx, py, px, nch, mch);
     } else {
+        const long tiles2 = (long)(K / 64) * (C / 64) * R * S;
+        const int mch = pick_mch(M, tiles2, 768);
+        const int nch = (int)((M + mch - 1) / mch);
         dim3 grid((unsigned)(K / 64), (unsigned)(C / 64),
                   (unsigned)(R * S * nch));
         const size_t shmem = 3 * 2 * (WG_BM * 64) * sizeof(unsigned short);
         hipLaunchKernelGGL(conv_wgrad_kernel<2>, grid, dim3(256), shmem,
                            stream, (const unsigned short*)dy,
                            (const unsigned short*)x, ws, N, H, W, C, K, P, Q,
-                           R, S, sy, sx, py, px, nch, WG_MCH);
+                           R, S, sy, sx, py, px, nch, mch);
     }
 }
 

@@ -178,11 +178,10 @@ class _BNAct(torch.autograd.Function):
         gw_sl = flat_grad_slice(weight)
         gb_sl = flat_grad_slice(bias) if bias is not None else None
         direct = gw_sl is not None and gb_sl is not None
-        empty = torch.empty(0, device=x.device, dtype=torch.float32)
         gx, gw, gb = C.bn_act_bwd(
             grad_out.contiguous(memory_format=torch.channels_last),
             x, weight, save_mean, save_invstd, out, ctx.relu, ctx.training,
-            gw_sl if direct else empty, gb_sl if direct else empty,
+            gw_sl if direct else None, gb_sl if direct else None,
         )
         if direct:
             from ..parallel.bucketing import notify_grad_written
