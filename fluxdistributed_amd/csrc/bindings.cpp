@@ -119,10 +119,11 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
     return {out, save_mean, save_invstd};
 }
 
-std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
     at::Tensor gout, at::Tensor x, at::Tensor weight, at::Tensor save_mean,
     at::Tensor save_invstd, at::Tensor out, bool relu, bool training,
-    c10::optional<at::Tensor> gw_out, c10::optional<at::Tensor> gb_out) {
+    c10::optional<at::Tensor> gw_out, c10::optional<at::Tensor> gb_out,
+    bool want_gres) {
     // gw_out/gb_out set: fp32 flat-G slices, written += (direct grad)
     auto [rows, C] = nhwc_rows(x);
     auto gc = gout.contiguous(at::MemoryFormat::ChannelsLast);
@@ -143,13 +144,16 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
                              part.data_ptr<float>(), gw.data_ptr<float>(),
                              gb.data_ptr<float>(), rows, (int)C, relu,
                              training, direct, dt_of(x), stream);
+    at::Tensor gres;
+    if (want_gres) gres = at::empty_like(x);
     fda::bn_bwd_apply_launch(gc.data_ptr(), x.data_ptr(), out.data_ptr(),
                              save_mean.data_ptr<float>(),
                              save_invstd.data_ptr<float>(),
                              weight.data_ptr<float>(), ws.data_ptr<float>(),
-                             gx.data_ptr(), rows, (int)C, relu, training,
-                             dt_of(x), stream);
-    return {gx, gw, gb};
+                             gx.data_ptr(),
+                             want_gres ? gres.data_ptr() : nullptr,
+                             rows, (int)C, relu, training, dt_of(x), stream);
+    return {gx, gw, gb, gres};
 }
 
 std::tuple<at::Tensor, at::Tensor> maxpool_fwd(at::Tensor x, int64_t KH,
@@ -374,7 +378,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("out"), pybind11::arg("relu"),
           pybind11::arg("training"),
           pybind11::arg("gw_out") = pybind11::none(),
-          pybind11::arg("gb_out") = pybind11::none());
+          pybind11::arg("gb_out") = pybind11::none(),
+          pybind11::arg("want_gres") = false);
     m.def("maxpool_fwd", &maxpool_fwd);
     m.def("maxpool_bwd", &maxpool_bwd);
     m.def("sgd_step", &sgd_step);

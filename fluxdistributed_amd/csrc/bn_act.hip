@@ -281,12 +281,17 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ gout,
                                     const float* __restrict__ weight,
                                     const float* __restrict__ k1,
                                     const float* __restrict__ k2,
-                                    T* __restrict__ gx, int64_t nvec, int C) {
+                                    T* __restrict__ gx,
+                                    T* __restrict__ gres,  // may be null
+                                    int64_t nvec, int C) {
+    // gres: the residual-branch gradient relu_mask*gout — a byproduct of
+    // this kernel's own mask computation (saves the separate add_relu_bwd
+    // pass the residual path used to run).
     const int cvec = C / V;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
          i += (int64_t)gridDim.x * blockDim.x) {
         const int c0 = (int)(i % cvec) * V;
-        T gv[V], xv[V], ov[V], rv_[V];
+        T gv[V], xv[V], ov[V], rv_[V], mv[V];
         *(uint4*)gv = ((const uint4*)gout)[i];
         *(uint4*)xv = ((const uint4*)x)[i];
         if constexpr (RELU) *(uint4*)ov = ((const uint4*)out)[i];
@@ -295,12 +300,14 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ gout,
             const int c = c0 + k;
             float g = load_f32(gv + k);
             if constexpr (RELU) g = load_f32(ov + k) > 0.f ? g : 0.f;
+            store_f32(mv + k, g);
             const float is = invstd[c];
             const float xhat = (load_f32(xv + k) - mean[c]) * is;
             const float r = (g - k1[c] - xhat * k2[c]) * weight[c] * is;
             store_f32(rv_ + k, r);
         }
         ((uint4*)gx)[i] = *(uint4*)rv_;
+        if (gres != nullptr) ((uint4*)gres)[i] = *(uint4*)mv;
     }
 }
 
@@ -415,8 +422,8 @@ void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
 void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
                          const float* save_mean, const float* save_invstd,
                          const float* weight, const float* ws, void* gx,
-                         int64_t rows, int C, bool relu, bool /*training*/,
-                         DT dt, hipStream_t s) {
+                         void* gres, int64_t rows, int C, bool relu,
+                         bool /*training*/, DT dt, hipStream_t s) {
     const int V = dt == DT::BF16 ? 8 : 4;
     const int64_t nvec = rows * C / V;
     int64_t blocks = (nvec + 255) / 256;
@@ -427,7 +434,7 @@ void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
         hipLaunchKernelGGL((bn_bwd_apply_kernel<T, VW, RELU_>), dim3(grid),     \
                            dim3(256), 0, s, (const T*)gout, (const T*)x,        \
                            (const T*)out, save_mean, save_invstd, weight, k1,   \
-                           k2, (T*)gx, nvec, C)
+                           k2, (T*)gx, (T*)gres, nvec, C)
     if (dt == DT::BF16) { if (relu) FDA_BAPPLY(unsigned short, 8, true); else FDA_BAPPLY(unsigned short, 8, false); }
     else { if (relu) FDA_BAPPLY(float, 4, true); else FDA_BAPPLY(float, 4, false); }
     #undef FDA_BAPPLY

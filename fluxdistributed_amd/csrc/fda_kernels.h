@@ -52,8 +52,8 @@ void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
 void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
                          const float* save_mean, const float* save_invstd,
                          const float* weight, const float* ws, void* gx,
-                         int64_t rows, int C, bool relu, bool training, DT dt,
-                         hipStream_t s);
+                         void* gres, int64_t rows, int C, bool relu,
+                         bool training, DT dt, hipStream_t s);
 
 // MaxPool2d NHWC with saved argmax byte per output element
 void maxpool_fwd_launch(const void* x, void* out, uint8_t* idx, int N, int H,

@@ -178,10 +178,11 @@ class _BNAct(torch.autograd.Function):
         gw_sl = flat_grad_slice(weight)
         gb_sl = flat_grad_slice(bias) if bias is not None else None
         direct = gw_sl is not None and gb_sl is not None
-        gx, gw, gb = C.bn_act_bwd(
+        want_gres = ctx.has_residual and ctx.relu
+        gx, gw, gb, gres_k = C.bn_act_bwd(
             grad_out.contiguous(memory_format=torch.channels_last),
             x, weight, save_mean, save_invstd, out, ctx.relu, ctx.training,
-            gw_sl if direct else None, gb_sl if direct else None,
+            gw_sl if direct else None, gb_sl if direct else None, want_gres,
         )
         if direct:
             from ..parallel.bucketing import notify_grad_written
@@ -189,13 +190,11 @@ class _BNAct(torch.autograd.Function):
             notify_grad_written(weight)
             notify_grad_written(bias)
             gw = gb = None
-        gres = gx_res = None
+        gres = None
         if ctx.has_residual:
-            # d(out)/d(residual) = relu-mask * grad_out = first stage of gx;
-            # the kernel returns it as gx's sibling (same mask, no bn chain).
-            gx_res = C.add_relu_bwd(grad_out.contiguous(memory_format=torch.channels_last), out) \
-                if ctx.relu else grad_out
-            gres = gx_res
+            # d(out)/d(residual) = relu-mask * grad_out: a byproduct of the
+            # bwd-apply kernel's own mask computation (gres output).
+            gres = gres_k if ctx.relu else grad_out
         return gx, gw, gb, None, None, None, None, None, None, gres
 
 

@@ -96,7 +96,7 @@ def test_bn_act_bwd(dtype, tol, Cc, relu, seed):
     out, mean, invstd = C.bn_act_fwd(x, w, b, rm, rv, True, 0.1, 1e-5, relu,
                                      torch.empty(0, device=DEV, dtype=dtype))
     gout = torch.randn_like(x).contiguous(memory_format=torch.channels_last)
-    gx, gw, gb = C.bn_act_bwd(gout, x, w, mean, invstd, out, relu, True)
+    gx, gw, gb, _ = C.bn_act_bwd(gout, x, w, mean, invstd, out, relu, True)
 
     # fp32 autograd reference
     xr = x.float().detach().requires_grad_()
