@@ -21,20 +21,33 @@ import torch
 
 
 class EagerTrainStep:
-    def __init__(self, model, optimizer, loss_fn: Callable, ddp=None):
+    """Eager step with rocprof-visible stage markers: torch.cuda.nvtx maps
+    to roctx ranges on ROCm, so `rocprofv3 --marker-trace` shows
+    fwd/bwd/allreduce/step spans (observability the reference lacks,
+    SURVEY.md §5.1)."""
+
+    def __init__(self, model, optimizer, loss_fn: Callable, ddp=None,
+                 markers: bool = False):
         self.model = model
         self.optimizer = optimizer
         self.loss_fn = loss_fn
         self.ddp = ddp
+        self.markers = markers and torch.cuda.is_available()
 
     def __call__(self, x, y):
+        nvtx = torch.cuda.nvtx if self.markers else None
+        if nvtx: nvtx.range_push("fwd")
         out = self.model(x)
         loss = self.loss_fn(out, y)
+        if nvtx: nvtx.range_pop(); nvtx.range_push("bwd")
         self.optimizer.zero_grad()
         loss.backward()
+        if nvtx: nvtx.range_pop(); nvtx.range_push("allreduce")
         if self.ddp is not None:
             self.ddp.finalize_backward()
+        if nvtx: nvtx.range_pop(); nvtx.range_push("step")
         self.optimizer.step()
+        if nvtx: nvtx.range_pop()
         return loss
 
 
@@ -89,7 +102,8 @@ class GraphedTrainStep:
 
 
 def make_train_step(model, optimizer, loss_fn, example_batch=None, ddp=None,
-                    use_graph: bool = True, warmup: int = 3):
+                    use_graph: bool = True, warmup: int = 3,
+                    markers: bool = False):
     """Pick graph or eager. Graph requires CUDA + an example batch; falls
     back to eager if capture fails (e.g. an op is not capture-safe)."""
     if use_graph and example_batch is not None and example_batch[0].is_cuda:
@@ -101,4 +115,5 @@ def make_train_step(model, optimizer, loss_fn, example_batch=None, ddp=None,
             import warnings
 
             warnings.warn(f"hipGraph capture failed ({e}); using eager step")
-    return EagerTrainStep(model, optimizer, loss_fn, ddp=ddp)
+    return EagerTrainStep(model, optimizer, loss_fn, ddp=ddp,
+                          markers=markers)
