@@ -243,8 +243,10 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
                     acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         af[mi][kh], bf[ni][kh], acc[mi][ni], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
+        // no end-of-iteration barrier: the next iteration's vmcnt + top
+        // barrier already orders buffer reuse (a wave reaches that barrier
+        // only after its lgkm-waited fragment reads of this buffer), and
+        // the final iteration needs no sync before the epilogue.
     }
 
     // ---- epilogue: C/D map col=lane&15, row=(lane>>4)*4+j ----------------
@@ -569,10 +571,8 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
                         af, bf, acc[ki][ci], 0, 0, 0);
                 }
         __builtin_amdgcn_s_setprio(0);
-        if (RING == 2) {
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        }
-        __builtin_amdgcn_s_barrier();
+        // end barrier dropped: the next iteration's wait + top barrier
+        // orders buffer reuse for every ring depth.
     }
 
     // epilogue: out[i=k][j=c]; C/D map col=lane&15, row=(lane>>4)*4+jj
