@@ -217,9 +217,13 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     // ---- main loop: double buffer; tile t+1's DMA in flight over tile t's
     // compute, drained at the iteration boundary (guide T3 minimum form) ---
     if (T > 0) stage(0, 0);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
     for (int it = 0; it < T; ++it) {
+        // tile `it` landed chip-wide: each wave drains its own DMA, the
+        // barrier joins all waves. This is the loop's ONLY barrier — the
+        // next K-step's staging targets the buffer every wave finished
+        // reading before it arrived here.
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
         if (it + 1 < T) stage((it + 1) & 1, it + 1);
         const unsigned short* buf = lds + (it & 1) * BUF_ELEMS;
         short8 af[4][2], bf[4][2];
