@@ -34,6 +34,7 @@
 // stem (C=3) falls back to the library path.
 
 #include <hip/hip_runtime.h>
+#include <cstdlib>
 #include "fda_common.h"
 
 namespace fda {
@@ -56,7 +57,7 @@ enum ConvMode { CONV_FWD = 0, CONV_DGRAD = 1, CONV_STEM = 2 };
 
 constexpr int BK = 64;
 
-template <int MODE, int BM, int BN, int WN>
+template <int MODE, int BM, int BN, int WN, int NBUF = 2>
 __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     const unsigned short* __restrict__ src,
     const unsigned short* __restrict__ wgt,
@@ -98,7 +99,8 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     const int wm = wid / WN;             // wave row (64-pixel granularity)
     const int wn = wid % WN;             // wave col (64-channel granularity)
 
-    extern __shared__ unsigned short lds[];   // [2][BUF_ELEMS]
+    extern __shared__ unsigned short lds[];   // [NBUF][BUF_ELEMS]
+    constexpr int GPW = BM / 32 + BN / 32;    // glds per wave per tile
 
     // ---- per-lane staging descriptors ------------------------------------
     int a_row[AI];
@@ -217,15 +219,24 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     // ---- main loop: double buffer; tile t+1's DMA in flight over tile t's
     // compute, drained at the iteration boundary (guide T3 minimum form) ---
     if (T > 0) stage(0, 0);
+    if (NBUF > 2 && T > 1) stage(1, 1);
     for (int it = 0; it < T; ++it) {
         // tile `it` landed chip-wide: each wave drains its own DMA, the
         // barrier joins all waves. This is the loop's ONLY barrier — the
         // next K-step's staging targets the buffer every wave finished
-        // reading before it arrived here.
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        // reading before it arrived here. With NBUF==3 the counted wait
+        // leaves tile it+1's DMA in flight across the barrier (T4).
+        if (NBUF > 2 && it + 1 < T) {
+            if (GPW == 8)
+                asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+        } else {
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
         __builtin_amdgcn_s_barrier();
-        if (it + 1 < T) stage((it + 1) & 1, it + 1);
-        const unsigned short* buf = lds + (it & 1) * BUF_ELEMS;
+        if (it + NBUF - 1 < T) stage((it + NBUF - 1) % NBUF, it + NBUF - 1);
+        const unsigned short* buf = lds + (it % NBUF) * BUF_ELEMS;
         short8 af[4][2], bf[4][2];
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
@@ -279,7 +290,7 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     }
 }
 
-template <int MODE, int BM, int BN, int WN>
+template <int MODE, int BM, int BN, int WN, int NBUF = 2>
 static void launch_cfg(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
@@ -290,8 +301,17 @@ static void launch_cfg(const void* src, const void* wgt, void* out,
         : (long)N * ((H + sy - 1) / sy) * ((W + sx - 1) / sx);
     dim3 grid((unsigned)((M + BM - 1) / BM), (unsigned)(OC / BN),
               (MODE != CONV_DGRAD) ? 1u : (unsigned)(sy * sx));
-    const size_t shmem = 2 * (BM * BK + BN * BK) * sizeof(unsigned short);
-    hipLaunchKernelGGL((conv_igemm_kernel<MODE, BM, BN, WN>), grid, dim3(256),
+    const size_t shmem = NBUF * (BM * BK + BN * BK) * sizeof(unsigned short);
+    if (shmem > 65536) {
+        static bool raised = [] {
+            hipFuncSetAttribute(
+                (const void*)&conv_igemm_kernel<MODE, BM, BN, WN, NBUF>,
+                hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+            return true;
+        }();
+        (void)raised;
+    }
+    hipLaunchKernelGGL((conv_igemm_kernel<MODE, BM, BN, WN, NBUF>), grid, dim3(256),
                        shmem, stream, (const unsigned short*)src,
                        (const unsigned short*)wgt, (unsigned short*)out,
                        N, H, W, C, K, P, Q, R, S, sy, sx, py, px);
@@ -306,13 +326,29 @@ void conv_stem_fwd_launch(const void* src, const void* wgt, void* out,
                                       P, Q, R, /*S=*/1, sy, sx, 0, 0, stream);
 }
 
+static int conv_nbuf() {
+    // A/B knob: FLUXDIST_CONV_NBUF=3 -> 3-buffer counted ring (1 block/CU
+    // for the 128x128 config; 96 KiB LDS) vs default 2 (2 blocks/CU).
+    static int v = [] {
+        const char* e = getenv("FLUXDIST_CONV_NBUF");
+        return (e && e[0] == '3') ? 3 : 2;
+    }();
+    return v;
+}
+
 void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
                        bool dgrad, hipStream_t stream) {
     const int OC = dgrad ? C : K;
+    const bool big = OC % 128 == 0;
+    const bool nb3 = conv_nbuf() == 3;
     if (dgrad) {
-        if (OC % 128 == 0)
+        if (big && nb3)
+            launch_cfg<CONV_DGRAD, 128, 128, 2, 3>(src, wgt, out, N, H, W, C,
+                                                   K, P, Q, R, S, sy, sx, py,
+                                                   px, stream);
+        else if (big)
             launch_cfg<CONV_DGRAD, 128, 128, 2>(src, wgt, out, N, H, W, C, K,
                                                 P, Q, R, S, sy, sx, py, px,
                                                 stream);
@@ -321,7 +357,11 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                                                P, Q, R, S, sy, sx, py, px,
                                                stream);
     } else {
-        if (OC % 128 == 0)
+        if (big && nb3)
+            launch_cfg<CONV_FWD, 128, 128, 2, 3>(src, wgt, out, N, H, W, C,
+                                                 K, P, Q, R, S, sy, sx, py,
+                                                 px, stream);
+        else if (big)
             launch_cfg<CONV_FWD, 128, 128, 2>(src, wgt, out, N, H, W, C, K,
                                               P, Q, R, S, sy, sx, py, px,
                                               stream);
