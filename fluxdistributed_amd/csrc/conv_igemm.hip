@@ -140,51 +140,69 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     const int cblocks = RC / BK;
     const int T = nR * nS * cblocks;
 
+    // Incremental staging offsets: stage() is called with strictly
+    // increasing `it` (prologue, then it+NBUF-1), so within one filter tap
+    // the source offsets just advance by BK; the full per-row address math
+    // (multiply + bounds check) runs only on tap changes — it was
+    // comparable to the whole MFMA phase per K-step otherwise.
+    long a_goff[AI];
+    bool a_okc[AI];
+    long b_goff[BI];
+    int last_rsi = -1;
     auto stage = [&](int buf, int it) {
         const int rsi = it / cblocks;
-        const int cb = (it % cblocks) * BK;
         const int ri = rsi / nS, si = rsi % nS;
         unsigned short* base = lds + buf * BUF_ELEMS;
+        if (rsi != last_rsi) {
+            last_rsi = rsi;
+            const int cb = (it % cblocks) * BK;   // 0 except NBUF>1 prologue
+            #pragma unroll
+            for (int i = 0; i < AI; ++i) {
+                const int cs = (cslot ^ (a_row[i] & 7)) * 8;
+                bool ok = a_mok[i];
+                long off = 0;
+                if (MODE == CONV_STEM) {
+                    off = a_pix[i] + ((long)(a_hb[i] + ri) * W + a_wb[i]) * 8 + cs;
+                } else if (MODE == CONV_FWD) {
+                    const int h = a_hb[i] + ri, w = a_wb[i] + si;
+                    ok = ok && (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W;
+                    off = a_pix[i] + ((long)h * W + w) * C + cb + cs;
+                } else {
+                    const int p = a_hb[i] - ri, q = a_wb[i] - si;
+                    ok = ok && (unsigned)p < (unsigned)P && (unsigned)q < (unsigned)Q;
+                    off = a_pix[i] + ((long)p * Q + q) * K + cb + cs;
+                }
+                a_goff[i] = off;
+                a_okc[i] = ok;
+            }
+            const int r = r0 + ri * ((MODE == CONV_FWD) ? 1 : sy);
+            const int s = s0 + si * ((MODE == CONV_FWD) ? 1 : sx);
+            const int rs = r * S + s;
+            #pragma unroll
+            for (int i = 0; i < BI; ++i) {
+                const int row = b_row[i];
+                const int cs = (cslot ^ (row & 7)) * 8;
+                if (MODE == CONV_STEM)
+                    b_goff[i] = ((long)(n0 + row) * R + ri) * 64 + cs;
+                else if (MODE == CONV_FWD)
+                    b_goff[i] = (long)(n0 + row) * R * S * C + (long)rs * C + cb + cs;
+                else
+                    b_goff[i] = (long)((long)rs * C + n0 + row) * K + cb + cs;
+            }
+        } else {
+            #pragma unroll
+            for (int i = 0; i < AI; ++i) a_goff[i] += BK;
+            #pragma unroll
+            for (int i = 0; i < BI; ++i) b_goff[i] += BK;
+        }
         #pragma unroll
         for (int i = 0; i < AI; ++i) {
-            const int row = a_row[i];
-            const int cs = (cslot ^ (row & 7)) * 8;
-            const unsigned short* sp;
-            bool ok = a_mok[i];
-            long off = 0;
-            if (MODE == CONV_STEM) {
-                // 64 virtual channels of K-step ri = pixels (h, w..w+7)x8ch,
-                // all in-bounds on the padded image
-                off = a_pix[i] + ((long)(a_hb[i] + ri) * W + a_wb[i]) * 8 + cs;
-            } else if (MODE == CONV_FWD) {
-                const int h = a_hb[i] + ri, w = a_wb[i] + si;
-                ok = ok && (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W;
-                off = a_pix[i] + ((long)h * W + w) * C + cb + cs;
-            } else {
-                const int p = a_hb[i] - ri, q = a_wb[i] - si;
-                ok = ok && (unsigned)p < (unsigned)P && (unsigned)q < (unsigned)Q;
-                off = a_pix[i] + ((long)p * Q + q) * K + cb + cs;
-            }
-            sp = ok ? src + off : conv_zero16;
+            const unsigned short* sp = a_okc[i] ? src + a_goff[i] : conv_zero16;
             FDA_GLDS16(sp, base + (wid * AI + i) * 8 * BK);
         }
-        const int r = r0 + ri * ((MODE == CONV_FWD) ? 1 : sy);
-        const int s = s0 + si * ((MODE == CONV_FWD) ? 1 : sx);
-        const int rs = r * S + s;
         #pragma unroll
         for (int i = 0; i < BI; ++i) {
-            const int row = b_row[i];
-            const int cs = (cslot ^ (row & 7)) * 8;
-            const unsigned short* sp;
-            if (MODE == CONV_STEM) {
-                // wpad[k][r][64]
-                sp = wgt + ((long)(n0 + row) * R + ri) * 64 + cs;
-            } else if (MODE == CONV_FWD) {
-                sp = wgt + ((long)(n0 + row) * R * S * C + (long)rs * C + cb + cs);
-            } else {
-                sp = wgt + ((long)((long)rs * C + n0 + row) * K + cb + cs);
-            }
-            FDA_GLDS16(sp, base + A_ELEMS + (wid * BI + i) * 8 * BK);
+            FDA_GLDS16(wgt + b_goff[i], base + A_ELEMS + (wid * BI + i) * 8 * BK);
         }
     };
 
