@@ -74,7 +74,7 @@ at::Tensor add_relu_bwd(at::Tensor gout, at::Tensor out) {
 std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
     at::Tensor x, at::Tensor weight, at::Tensor bias, at::Tensor running_mean,
     at::Tensor running_var, bool training, double momentum, double eps,
-    bool relu, at::Tensor residual) {
+    bool relu, at::Tensor residual, c10::optional<at::Tensor> conv_part) {
     auto [rows, C] = nhwc_rows(x);
     const int V = dt_of(x) == DT::BF16 ? 8 : 4;
     TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
@@ -92,7 +92,17 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
     auto out = at::empty_like(x);
     auto stream = cur_stream();
 
-    if (training) {
+    if (training && conv_part.has_value()) {
+        // stats already accumulated by the producing conv's epilogue
+        auto& cp = *conv_part;
+        TORCH_CHECK(cp.dim() == 3 && cp.size(1) == 2 && cp.size(2) == C);
+        fda::bn_finalize_from_partials_launch(
+            cp.data_ptr<float>(), (int)cp.size(0), weight.data_ptr<float>(),
+            bias.data_ptr<float>(), running_mean.data_ptr<float>(),
+            running_var.data_ptr<float>(), save_mean.data_ptr<float>(),
+            save_invstd.data_ptr<float>(), ws.data_ptr<float>(), rows,
+            (int)C, (float)momentum, (float)eps, stream);
+    } else if (training) {
         TORCH_CHECK(C % 64 == 0, "training BN needs C % 64 == 0, got ", C);
         auto part = at::empty({fda::bn_stats_partial_floats((int)C, rows, dt_of(x))},
                               fopts);
@@ -247,6 +257,34 @@ at::Tensor conv_igemm_fwd(at::Tensor x, at::Tensor w,
     return y;
 }
 
+std::tuple<at::Tensor, at::Tensor> conv_igemm_fwd_stats(
+    at::Tensor x, at::Tensor w, int64_t sy, int64_t sx, int64_t py,
+    int64_t px) {
+    // forward + per-m-tile BN partials [mtiles][2][K] (sum/sumsq of the
+    // rounded output) — feeds bn_finalize_from_partials.
+    TORCH_CHECK(x.dim() == 4 && w.dim() == 4);
+    TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+                x.is_contiguous(at::MemoryFormat::ChannelsLast));
+    TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast));
+    const int N = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+              W = (int)x.size(3);
+    const int K = (int)w.size(0), R = (int)w.size(2), S = (int)w.size(3);
+    TORCH_CHECK(C % 64 == 0 && K % 64 == 0);
+    const int P = (H + 2 * (int)py - R) / (int)sy + 1;
+    const int Q = (W + 2 * (int)px - S) / (int)sx + 1;
+    const long M = (long)N * P * Q;
+    const int BM = (K % 128 == 0) ? 128 : 256;
+    const long mtiles = (M + BM - 1) / BM;
+    auto y = at::empty({N, K, P, Q},
+                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    auto part = at::empty({mtiles, 2, K}, x.options().dtype(at::kFloat));
+    fda::conv_igemm_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                           N, H, W, C, K, P, Q, R, S, (int)sy, (int)sx,
+                           (int)py, (int)px, /*dgrad=*/false, cur_stream(),
+                           part.data_ptr<float>());
+    return {y, part};
+}
+
 at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
                             int64_t C, int64_t H, int64_t W,
                             int64_t R, int64_t S,
@@ -332,6 +370,24 @@ at::Tensor conv_stem_fwd(at::Tensor x8, at::Tensor wpad, int64_t R,
     return y;
 }
 
+std::tuple<at::Tensor, at::Tensor> conv_stem_fwd_stats(
+    at::Tensor x8, at::Tensor wpad, int64_t R, int64_t sy, int64_t sx,
+    int64_t P, int64_t Q) {
+    TORCH_CHECK(x8.size(1) == 8 &&
+                x8.is_contiguous(at::MemoryFormat::ChannelsLast));
+    const int N = (int)x8.size(0), Hp = (int)x8.size(2), Wp = (int)x8.size(3);
+    const int K = (int)wpad.size(0);
+    const long M = (long)N * P * Q;
+    const long mtiles = (M + 256 - 1) / 256;
+    auto y = at::empty({N, K, P, Q},
+                       x8.options().memory_format(at::MemoryFormat::ChannelsLast));
+    auto part = at::empty({mtiles, 2, K}, x8.options().dtype(at::kFloat));
+    fda::conv_stem_fwd_launch(x8.data_ptr(), wpad.data_ptr(), y.data_ptr(),
+                              N, Hp, Wp, K, (int)P, (int)Q, (int)R, (int)sy,
+                              (int)sx, cur_stream(), part.data_ptr<float>());
+    return {y, part};
+}
+
 at::Tensor conv_stem_wgrad(at::Tensor dy, at::Tensor x8, int64_t R,
                            int64_t sy, int64_t sx) {
     TORCH_CHECK(dy.scalar_type() == at::kBFloat16 &&
@@ -367,11 +423,37 @@ void wt_transpose_batch(at::Tensor src_ptrs, at::Tensor dst_ptrs,
                              cur_stream());
 }
 
+at::Tensor gap_fwd(at::Tensor x) {
+    auto [rows, C] = nhwc_rows(x);
+    const int N = (int)x.size(0);
+    const int HW = (int)(rows / N);
+    auto y = at::empty({N, C}, x.options());
+    fda::gap_fwd_launch(x.data_ptr(), y.data_ptr(), N, HW, (int)C, dt_of(x),
+                        cur_stream());
+    return y;
+}
+
+at::Tensor gap_bwd(at::Tensor gy, int64_t H, int64_t W) {
+    TORCH_CHECK(gy.dim() == 2 && gy.is_contiguous());
+    const int N = (int)gy.size(0), C = (int)gy.size(1);
+    auto gx = at::empty({N, C, H, W},
+                        gy.options().memory_format(at::MemoryFormat::ChannelsLast));
+    fda::gap_bwd_launch(gy.data_ptr(), gx.data_ptr(), N, (int)(H * W), C,
+                        dt_of(gy), cur_stream());
+    return gx;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ce_fwd", &ce_fwd, "fused logit cross-entropy fwd (loss + dlogits)");
     m.def("add_relu_fwd", &add_relu_fwd);
     m.def("add_relu_bwd", &add_relu_bwd);
-    m.def("bn_act_fwd", &bn_act_fwd);
+    m.def("bn_act_fwd", &bn_act_fwd, pybind11::arg("x"),
+          pybind11::arg("weight"), pybind11::arg("bias"),
+          pybind11::arg("running_mean"), pybind11::arg("running_var"),
+          pybind11::arg("training"), pybind11::arg("momentum"),
+          pybind11::arg("eps"), pybind11::arg("relu"),
+          pybind11::arg("residual"),
+          pybind11::arg("conv_part") = pybind11::none());
     m.def("bn_act_bwd", &bn_act_bwd, pybind11::arg("gout"),
           pybind11::arg("x"), pybind11::arg("weight"),
           pybind11::arg("save_mean"), pybind11::arg("save_invstd"),
@@ -382,8 +464,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("want_gres") = false);
     m.def("maxpool_fwd", &maxpool_fwd);
     m.def("maxpool_bwd", &maxpool_bwd);
+    m.def("gap_fwd", &gap_fwd);
+    m.def("gap_bwd", &gap_bwd);
     m.def("sgd_step", &sgd_step);
     m.def("adam_step", &adam_step);
+    m.def("conv_igemm_fwd_stats", &conv_igemm_fwd_stats);
+    m.def("conv_stem_fwd_stats", &conv_stem_fwd_stats);
     m.def("conv_igemm_fwd", &conv_igemm_fwd,
           "implicit-GEMM conv fwd (NHWC bf16, MFMA)");
     m.def("wt_transpose_batch", &wt_transpose_batch);

@@ -419,6 +419,18 @@ void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
     }
 }
 
+void bn_finalize_from_partials_launch(
+    const float* part, int NB, const float* weight, const float* bias,
+    float* rm, float* rv, float* save_mean, float* save_invstd, float* ws,
+    int64_t rows, int C, float momentum, float eps, hipStream_t s) {
+    // partials produced by the conv epilogue ([NB][2][C], single chunk);
+    // same reduce+finalize kernel the in-house stats pass feeds.
+    hipLaunchKernelGGL(bn_fwd_reduce_finalize_kernel, dim3(C / 64),
+                       dim3(256), 0, s, part, weight, bias, rm, rv,
+                       save_mean, save_invstd, ws, NB, C, 0, C, rows,
+                       momentum, eps);
+}
+
 void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
                          const float* save_mean, const float* save_invstd,
                          const float* weight, const float* ws, void* gx,

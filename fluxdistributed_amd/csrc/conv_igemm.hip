@@ -64,7 +64,11 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     unsigned short* __restrict__ out,
     int N, int H, int W, int C,
     int K, int P, int Q,
-    int R, int S, int sy, int sx, int py, int px) {
+    int R, int S, int sy, int sx, int py, int px,
+    float* __restrict__ stats /* [mtiles][2][OC] or null: per-channel
+                                 sum/sumsq of the rounded output — feeds the
+                                 BN reduce+finalize directly (the separate
+                                 bn_stats read pass is skipped) */) {
     constexpr int A_ELEMS = BM * BK;
     constexpr int B_ELEMS = BN * BK;
     constexpr int BUF_ELEMS = A_ELEMS + B_ELEMS;
@@ -284,6 +288,8 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
 
     // ---- epilogue: C/D map col=lane&15, row=(lane>>4)*4+j ----------------
     const int fcol = lane & 15, frow0 = (lane >> 4) * 4;
+    float ssum[4] = {0.f, 0.f, 0.f, 0.f};   // per-ni channel sums (rounded y)
+    float sq[4] = {0.f, 0.f, 0.f, 0.f};
     #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
         #pragma unroll
@@ -302,8 +308,50 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
             }
             unsigned short* orow = out + obase + n0 + wn * 64;
             #pragma unroll
-            for (int ni = 0; ni < 4; ++ni)
-                orow[ni * 16 + fcol] = f32_to_bf16bits(acc[mi][ni][j]);
+            for (int ni = 0; ni < 4; ++ni) {
+                const unsigned short us = f32_to_bf16bits(acc[mi][ni][j]);
+                orow[ni * 16 + fcol] = us;
+                if (MODE != CONV_DGRAD && stats != nullptr) {
+                    const float v = bf16bits_to_f32(us);
+                    ssum[ni] += v;
+                    sq[ni] += v * v;
+                }
+            }
+        }
+    }
+    if (MODE != CONV_DGRAD && stats != nullptr) {
+        // fold the four 16-row lane groups, then the waves sharing this
+        // channel column, then write this m-tile's [2][OC] partial slice
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+            #pragma unroll
+            for (int off = 16; off < 64; off <<= 1) {
+                ssum[ni] += __shfl_xor(ssum[ni], off, 64);
+                sq[ni] += __shfl_xor(sq[ni], off, 64);
+            }
+        }
+        float* sf = (float*)lds;             // staging LDS is free now
+        __builtin_amdgcn_s_barrier();
+        if ((lane >> 4) == 0) {
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni) {
+                sf[(wid * 4 + ni) * 16 + fcol] = ssum[ni];
+                sf[256 + (wid * 4 + ni) * 16 + fcol] = sq[ni];
+            }
+        }
+        __builtin_amdgcn_s_barrier();
+        if (wm == 0 && (lane >> 4) == 0) {
+            float* prow = stats + (long)blockIdx.x * 2 * OC + n0 + wn * 64;
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni) {
+                float s = ssum[ni], z = sq[ni];
+                for (int w2 = wid + WN; w2 < 4; w2 += WN) {
+                    s += sf[(w2 * 4 + ni) * 16 + fcol];
+                    z += sf[256 + (w2 * 4 + ni) * 16 + fcol];
+                }
+                prow[ni * 16 + fcol] = s;
+                prow[OC + ni * 16 + fcol] = z;
+            }
         }
     }
 }
@@ -312,7 +360,7 @@ template <int MODE, int BM, int BN, int WN, int NBUF = 2>
 static void launch_cfg(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
-                       hipStream_t stream) {
+                       hipStream_t stream, float* stats = nullptr) {
     const int OC = (MODE == CONV_DGRAD) ? C : K;
     const long M = (MODE != CONV_DGRAD)
         ? (long)N * P * Q
@@ -332,16 +380,18 @@ static void launch_cfg(const void* src, const void* wgt, void* out,
     hipLaunchKernelGGL((conv_igemm_kernel<MODE, BM, BN, WN, NBUF>), grid, dim3(256),
                        shmem, stream, (const unsigned short*)src,
                        (const unsigned short*)wgt, (unsigned short*)out,
-                       N, H, W, C, K, P, Q, R, S, sy, sx, py, px);
+                       N, H, W, C, K, P, Q, R, S, sy, sx, py, px, stats);
 }
 
 void conv_stem_fwd_launch(const void* src, const void* wgt, void* out,
                           int N, int Hp, int Wp, int K, int P, int Q,
-                          int R, int sy, int sx, hipStream_t stream) {
+                          int R, int sy, int sx, hipStream_t stream,
+                          float* stats) {
     // src: channel-padded (C=8) spatially pre-padded input [N,Hp,Wp,8];
     // wgt: wpad [K][R][64]; out: [N*P*Q][K]
     launch_cfg<CONV_STEM, 256, 64, 1>(src, wgt, out, N, Hp, Wp, /*C=*/8, K,
-                                      P, Q, R, /*S=*/1, sy, sx, 0, 0, stream);
+                                      P, Q, R, /*S=*/1, sy, sx, 0, 0, stream,
+                                      stats);
 }
 
 static int conv_nbuf() {
@@ -357,7 +407,7 @@ static int conv_nbuf() {
 void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
-                       bool dgrad, hipStream_t stream) {
+                       bool dgrad, hipStream_t stream, float* stats) {
     const int OC = dgrad ? C : K;
     const bool big = OC % 128 == 0;
     const bool nb3 = conv_nbuf() == 3;
@@ -378,15 +428,15 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
         if (big && nb3)
             launch_cfg<CONV_FWD, 128, 128, 2, 3>(src, wgt, out, N, H, W, C,
                                                  K, P, Q, R, S, sy, sx, py,
-                                                 px, stream);
+                                                 px, stream, stats);
         else if (big)
             launch_cfg<CONV_FWD, 128, 128, 2>(src, wgt, out, N, H, W, C, K,
                                               P, Q, R, S, sy, sx, py, px,
-                                              stream);
+                                              stream, stats);
         else
             launch_cfg<CONV_FWD, 256, 64, 1>(src, wgt, out, N, H, W, C, K,
                                              P, Q, R, S, sy, sx, py, px,
-                                             stream);
+                                             stream, stats);
     }
 }
 

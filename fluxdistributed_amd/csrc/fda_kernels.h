@@ -31,6 +31,12 @@ void bn_stats_launch(const void* x, float* ws, float* part,
                      float* running_mean, float* running_var, float* save_mean,
                      float* save_invstd, int64_t rows, int C, float momentum,
                      float eps, DT dt, hipStream_t s);
+// finalize from conv-epilogue partials ([NB][2][C])
+void bn_finalize_from_partials_launch(
+    const float* part, int NB, const float* weight, const float* bias,
+    float* rm, float* rv, float* save_mean, float* save_invstd, float* ws,
+    int64_t rows, int C, float momentum, float eps, hipStream_t s);
+
 // eval: scale/shift from running stats
 void bn_finalize_launch(float* ws, const float* weight, const float* bias,
                         float* running_mean, float* running_var,
@@ -63,18 +69,26 @@ void maxpool_bwd_launch(const void* gout, const uint8_t* idx, void* gx, int N,
                         int H, int W, int C, int HO, int WO, int KH, int KW,
                         int S, int P, DT dt, hipStream_t s);
 
+// global average pool (NHWC): y[n][c] = mean_hw x; gx = gy/HW broadcast
+void gap_fwd_launch(const void* x, void* y, int N, int HW, int C, DT dt,
+                    hipStream_t s);
+void gap_bwd_launch(const void* gy, void* gx, int N, int HW, int C, DT dt,
+                    hipStream_t s);
+
 // Implicit-GEMM conv, NHWC bf16 only (see conv_igemm.hip).
 // fwd:   src=x [N,H,W,C], wgt=w [K][R*S*C], out=y [N*P*Q][K]
 // dgrad: src=dy [N,P,Q,K], wgt=wt [R*S*C][K] (pre-transposed), out=dx
 void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
-                       bool dgrad, hipStream_t stream);
+                       bool dgrad, hipStream_t stream,
+                       float* stats = nullptr);
 
 // stem conv (small C via channel-pad to 8, spatially pre-padded input)
 void conv_stem_fwd_launch(const void* src, const void* wgt, void* out,
                           int N, int Hp, int Wp, int K, int P, int Q,
-                          int R, int sy, int sx, hipStream_t stream);
+                          int R, int sy, int sx, hipStream_t stream,
+                          float* stats = nullptr);
 void conv_stem_wgrad_launch(const void* dy, const void* x, float* ws,
                             int N, int Hp, int Wp, int K, int P, int Q,
                             int R, int sy, int sx, hipStream_t stream);

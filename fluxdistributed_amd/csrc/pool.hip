@@ -151,3 +151,84 @@ void maxpool_bwd_launch(const void* gout, const uint8_t* idx, void* gx, int N,
 }
 
 }  // namespace fda
+
+namespace fda {
+
+// ---- global average pool (AdaptiveMeanPool 1x1), NHWC -------------------
+// "AdaptiveMeanPool / global avg pool fwd/bwd — warp
+// reduction". fwd: y[n][c] = mean_hw x[n][h][w][c]; bwd: gx = gy/HW bcast.
+template <typename T, int V>
+__global__ __launch_bounds__(256) void gap_fwd_kernel(
+    const T* __restrict__ x, T* __restrict__ y, int N, int HW, int C) {
+    // one thread per (n, c-vec): strided column reduction, fp32 accum
+    const long nv = (long)N * (C / V);
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+         i += (long)gridDim.x * blockDim.x) {
+        const int n = (int)(i / (C / V));
+        const int c0 = (int)(i % (C / V)) * V;
+        const T* base = x + ((long)n * HW) * C + c0;
+        float acc[V];
+        #pragma unroll
+        for (int e = 0; e < V; ++e) acc[e] = 0.f;
+        for (int r = 0; r < HW; ++r) {
+            T v[V];
+            *(uint4*)v = *(const uint4*)(base + (long)r * C);
+            #pragma unroll
+            for (int e = 0; e < V; ++e) acc[e] += load_f32(v + e);
+        }
+        T o[V];
+        const float inv = 1.f / (float)HW;
+        #pragma unroll
+        for (int e = 0; e < V; ++e) store_f32(o + e, acc[e] * inv);
+        *(uint4*)(y + (long)n * C + c0) = *(uint4*)o;
+    }
+}
+
+template <typename T, int V>
+__global__ __launch_bounds__(256) void gap_bwd_kernel(
+    const T* __restrict__ gy, T* __restrict__ gx, int N, int HW, int C) {
+    const long nv = (long)N * HW * (C / V);
+    const float inv = 1.f / (float)HW;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+         i += (long)gridDim.x * blockDim.x) {
+        const long row = i / (C / V);          // (n, hw)
+        const int n = (int)(row / HW);
+        const int c0 = (int)(i % (C / V)) * V;
+        T g[V];
+        *(uint4*)g = *(const uint4*)(gy + (long)n * C + c0);
+        T o[V];
+        #pragma unroll
+        for (int e = 0; e < V; ++e) store_f32(o + e, load_f32(g + e) * inv);
+        *(uint4*)(gx + row * C + c0) = *(uint4*)o;
+    }
+}
+
+void gap_fwd_launch(const void* x, void* y, int N, int HW, int C, DT dt,
+                    hipStream_t s) {
+    const int V = dt == DT::BF16 ? 8 : 4;
+    const long nv = (long)N * (C / V);
+    dim3 grid((unsigned)((nv + 255) / 256 < 1024 ? (nv + 255) / 256 : 1024));
+    if (dt == DT::BF16)
+        hipLaunchKernelGGL((gap_fwd_kernel<unsigned short, 8>), grid,
+                           dim3(256), 0, s, (const unsigned short*)x,
+                           (unsigned short*)y, N, HW, C);
+    else
+        hipLaunchKernelGGL((gap_fwd_kernel<float, 4>), grid, dim3(256), 0, s,
+                           (const float*)x, (float*)y, N, HW, C);
+}
+
+void gap_bwd_launch(const void* gy, void* gx, int N, int HW, int C, DT dt,
+                    hipStream_t s) {
+    const int V = dt == DT::BF16 ? 8 : 4;
+    const long nv = (long)N * HW * (C / V);
+    dim3 grid((unsigned)((nv + 255) / 256 < 4096 ? (nv + 255) / 256 : 4096));
+    if (dt == DT::BF16)
+        hipLaunchKernelGGL((gap_bwd_kernel<unsigned short, 8>), grid,
+                           dim3(256), 0, s, (const unsigned short*)gy,
+                           (unsigned short*)gx, N, HW, C);
+    else
+        hipLaunchKernelGGL((gap_bwd_kernel<float, 4>), grid, dim3(256), 0, s,
+                           (const float*)gy, (float*)gx, N, HW, C);
+}
+
+}  // namespace fda

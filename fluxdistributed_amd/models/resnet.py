@@ -14,7 +14,8 @@ from typing import List, Optional, Type
 import torch
 import torch.nn as nn
 
-from ..ops.functional import batch_norm_act, fused_add_relu, MaxPool2d
+from ..ops.functional import (batch_norm_act, fused_add_relu, MaxPool2d,
+                              GlobalAvgPool)
 
 
 class FusedBNAct(nn.Module):
@@ -148,7 +149,7 @@ class ResNet(nn.Module):
         self.layer2 = self._make_layer(block, 128, layers[1], stride=2)
         self.layer3 = self._make_layer(block, 256, layers[2], stride=2)
         self.layer4 = self._make_layer(block, 512, layers[3], stride=2)
-        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.avgpool = GlobalAvgPool()
         self.fc = nn.Linear(512 * block.expansion, num_classes)
         self._init_weights()
 
@@ -172,7 +173,7 @@ class ResNet(nn.Module):
     def forward(self, x):
         x = self.maxpool(self.bn1(self.conv1(x)))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
-        x = self.avgpool(x).flatten(1)
+        x = self.avgpool(x)
         return self.fc(x)
 
 

@@ -16,6 +16,8 @@ from .functional import (  # noqa: F401
     batch_norm_act,
     max_pool2d,
     MaxPool2d,
+    global_avg_pool,
+    GlobalAvgPool,
 )
 from .fused_optim import FusedSGDMomentum, FusedAdam  # noqa: F401
 from .conv import fda_conv2d, FdaConv2d  # noqa: F401

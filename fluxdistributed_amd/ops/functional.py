@@ -153,10 +153,14 @@ class _BNAct(torch.autograd.Function):
         residual: Optional[torch.Tensor],
     ):
         C = require_native("batch_norm_act")
+        from .conv import take_conv_stats
+
+        conv_part = take_conv_stats(x) if training else None
         out, save_mean, save_invstd = C.bn_act_fwd(
             x, weight, bias, running_mean, running_var,
             training, momentum, eps, relu,
             residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype),
+            conv_part,
         )
         ctx.save_for_backward(x, weight, save_mean, save_invstd, out)
         ctx.bn_bias = bias
@@ -230,3 +234,35 @@ def batch_norm_act(
     if relu:
         out = torch.relu(out)
     return out.to(x.dtype)
+
+
+# --------------------------------------------------------------------------
+# Global average pool (the reference's AdaptiveMeanPool before the FC —
+# SURVEY.md §2.4 "warp reduction").
+# --------------------------------------------------------------------------
+
+
+class _GlobalAvgPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        C = require_native("global_avg_pool")
+        ctx.geom = (x.shape[2], x.shape[3])
+        return C.gap_fwd(x.contiguous(memory_format=torch.channels_last))
+
+    @staticmethod
+    def backward(ctx, gy):
+        C = require_native("global_avg_pool")
+        H, W = ctx.geom
+        return C.gap_bwd(gy.contiguous(), H, W)
+
+
+def global_avg_pool(x: torch.Tensor) -> torch.Tensor:
+    """[N,C,H,W] -> [N,C] mean over H,W."""
+    if _on_gpu(x):
+        return _GlobalAvgPool.apply(x)
+    return x.float().mean(dim=(2, 3)).to(x.dtype)
+
+
+class GlobalAvgPool(torch.nn.Module):
+    def forward(self, x):
+        return global_avg_pool(x)

@@ -193,3 +193,17 @@ def test_maxpool(dtype, tol, shape, k, s, p, seed):
     diff = (gx.float() - xr.grad).abs()
     frac_bad = float((diff > 1e-2).float().mean())
     assert frac_bad < 2e-3, frac_bad
+
+
+def test_global_avg_pool(seed):
+    from fluxdistributed_amd.ops.functional import global_avg_pool
+
+    x = torch.randn(6, 512, 7, 7, device=DEV, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    y = global_avg_pool(x)
+    ref = x.float().mean(dim=(2, 3))
+    assert (y.float() - ref).abs().max().item() < 2e-2
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    gref = (gy.float() / 49).unsqueeze(-1).unsqueeze(-1).expand(6, 512, 7, 7)
+    assert (x.grad.float() - gref).abs().max().item() < 1e-3
