@@ -5,8 +5,10 @@ The reference gets conv from cuDNN via NNlibCUDA; the
 MI355X-native path is fluxdistributed_amd/csrc/conv_igemm.hip —
 v_mfma_f32_16x16x32_bf16 tiles, LDS-staged via global_load_lds. Supported
 there: NHWC bf16, dilation 1, groups 1, C and K multiples of 64 (all
-ResNet body convs). The stem (C=3) and anything else goes to the library
-path (MIOpen through torch), and wgrad currently does too.
+ResNet body convs) — fwd, dgrad AND wgrad. Small-C stems (C<=5, e.g. the
+ImageNet 7x7 and CIFAR 3x3 stems) run on the dedicated CONV_STEM kernels;
+only genuinely unsupported shapes (grouped/dilated, odd channel counts)
+fall back to the library path.
 
 Env:
   FLUXDIST_CONV=miopen   force the library path everywhere (A/B testing)
