@@ -176,8 +176,8 @@ std::tuple<at::Tensor, at::Tensor> maxpool_fwd(at::Tensor x, int64_t KH,
     const int N = (int)x.size(0), H = (int)x.size(2), W = (int)x.size(3);
     const int HO = (int)((H + 2 * P - KH) / S + 1);
     const int WO = (int)((W + 2 * P - KW) / S + 1);
-    auto out = at::empty({N, C, HO, WO}, x.options())
-                   .contiguous(at::MemoryFormat::ChannelsLast);
+    auto out = at::empty({N, C, HO, WO},
+                         x.options().memory_format(at::MemoryFormat::ChannelsLast));
     auto idx = at::empty({(int64_t)N * HO * WO * C},
                          x.options().dtype(at::kByte));
     fda::maxpool_fwd_launch(x.data_ptr(), out.data_ptr(),
@@ -192,8 +192,8 @@ at::Tensor maxpool_bwd(at::Tensor gout, at::Tensor idx, int64_t H, int64_t W,
     auto gc = gout.contiguous(at::MemoryFormat::ChannelsLast);
     const int N = (int)gc.size(0), C = (int)gc.size(1);
     const int HO = (int)gc.size(2), WO = (int)gc.size(3);
-    auto gx = at::empty({N, C, (int)H, (int)W}, gc.options())
-                  .contiguous(at::MemoryFormat::ChannelsLast);
+    auto gx = at::empty({N, C, (int)H, (int)W},
+                        gc.options().memory_format(at::MemoryFormat::ChannelsLast));
     fda::maxpool_bwd_launch(gc.data_ptr(), idx.data_ptr<uint8_t>(),
                             gx.data_ptr(), N, (int)H, (int)W, C, HO, WO,
                             (int)KH, (int)KW, (int)S, (int)P, dt_of(gc),
