@@ -164,7 +164,8 @@ class _BNAct(torch.autograd.Function):
         )
         ctx.save_for_backward(x, weight, save_mean, save_invstd, out)
         ctx.bn_bias = bias
-        ctx.relu_mask = relu_mask if relu_mask.numel() else None
+        ctx.relu_mask = (relu_mask if relu_mask is not None and
+                         relu_mask.numel() else None)
         ctx.relu = relu
         ctx.has_residual = residual is not None
         ctx.training = training
