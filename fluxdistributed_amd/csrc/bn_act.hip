@@ -219,11 +219,15 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
 
 // ---- backward -------------------------------------------------------------
 
-template <typename T, int V, bool RELU>
+template <typename T, int V, bool RELU, bool MASKED>
 __global__ void bn_bwd_stats_kernel(const T* __restrict__ gout,
                                     const T* __restrict__ x,
                                     const T* __restrict__ out,
                                     const unsigned char* __restrict__ mask,
+                                    // MASKED is compile-time: a runtime
+                                    // mask!=null branch around the loads
+                                    // de-pipelines the streaming loop
+                                    // (guide §5 trap (c))
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
                                     float* __restrict__ part, int64_t rows,
@@ -248,21 +252,17 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ gout,
         *(uint4*)gv = *(const uint4*)(gout + r * C + c0);
         *(uint4*)xv = *(const uint4*)(x + r * C + c0);
         unsigned mbits = 0xffu;
-        if constexpr (RELU) {
-            if (mask != nullptr)
-                mbits = mask[(r * C + c0) / V];
-            else
-                *(uint4*)ov = *(const uint4*)(out + r * C + c0);
-        }
+        if constexpr (RELU && MASKED)
+            mbits = mask[(r * C + c0) / V];
+        else if constexpr (RELU)
+            *(uint4*)ov = *(const uint4*)(out + r * C + c0);
         #pragma unroll
         for (int k = 0; k < V; ++k) {
             float g = load_f32(gv + k);
-            if constexpr (RELU) {
-                if (mask != nullptr)
-                    g = ((mbits >> k) & 1u) ? g : 0.f;
-                else
-                    g = load_f32(ov + k) > 0.f ? g : 0.f;
-            }
+            if constexpr (RELU && MASKED)
+                g = ((mbits >> k) & 1u) ? g : 0.f;
+            else if constexpr (RELU)
+                g = load_f32(ov + k) > 0.f ? g : 0.f;
             const float xhat = (load_f32(xv + k) - mu[k]) * is[k];
             sg[k] += g;
             sgx[k] += g * xhat;
@@ -292,7 +292,7 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ gout,
     }
 }
 
-template <typename T, int V, bool RELU>
+template <typename T, int V, bool RELU, bool MASKED>
 __global__ void bn_bwd_apply_kernel(const T* __restrict__ gout,
                                     const T* __restrict__ x,
                                     const T* __restrict__ out,
@@ -316,22 +316,18 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ gout,
         *(uint4*)gv = ((const uint4*)gout)[i];
         *(uint4*)xv = ((const uint4*)x)[i];
         unsigned mbits = 0xffu;
-        if constexpr (RELU) {
-            if (mask != nullptr)
-                mbits = mask[i];
-            else
-                *(uint4*)ov = ((const uint4*)out)[i];
-        }
+        if constexpr (RELU && MASKED)
+            mbits = mask[i];
+        else if constexpr (RELU)
+            *(uint4*)ov = ((const uint4*)out)[i];
         #pragma unroll
         for (int k = 0; k < V; ++k) {
             const int c = c0 + k;
             float g = load_f32(gv + k);
-            if constexpr (RELU) {
-                if (mask != nullptr)
-                    g = ((mbits >> k) & 1u) ? g : 0.f;
-                else
-                    g = load_f32(ov + k) > 0.f ? g : 0.f;
-            }
+            if constexpr (RELU && MASKED)
+                g = ((mbits >> k) & 1u) ? g : 0.f;
+            else if constexpr (RELU)
+                g = load_f32(ov + k) > 0.f ? g : 0.f;
             store_f32(mv + k, g);
             const float is = invstd[c];
             const float xhat = (load_f32(xv + k) - mean[c]) * is;
@@ -438,7 +434,14 @@ void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
         const int c_base = ch * chunkC;
         const int cc = (C - c_base) < chunkC ? (C - c_base) : chunkC;
         #define FDA_BSTATS(T, VW, RELU_)                                        \
-            hipLaunchKernelGGL((bn_bwd_stats_kernel<T, VW, RELU_>), dim3(grid), \
+            if (mask != nullptr)                                                \
+                hipLaunchKernelGGL((bn_bwd_stats_kernel<T, VW, RELU_, true>),   \
+                                   dim3(grid), dim3(256), shmem, s,             \
+                                   (const T*)gout, (const T*)x, (const T*)out,  \
+                                   mask, save_mean, save_invstd, part, rows, C, \
+                                   c_base, cc);                                 \
+            else                                                                \
+                hipLaunchKernelGGL((bn_bwd_stats_kernel<T, VW, RELU_, false>), dim3(grid), \
                                dim3(256), shmem, s, (const T*)gout,             \
                                (const T*)x, (const T*)out, mask, save_mean,     \
                                save_invstd, part, rows, C, c_base, cc)
@@ -477,7 +480,14 @@ void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
     const float* k1 = ws + 2 * C;
     const float* k2 = ws + 3 * C;
     #define FDA_BAPPLY(T, VW, RELU_)                                            \
-        hipLaunchKernelGGL((bn_bwd_apply_kernel<T, VW, RELU_>), dim3(grid),     \
+        if (mask != nullptr)                                                    \
+            hipLaunchKernelGGL((bn_bwd_apply_kernel<T, VW, RELU_, true>),       \
+                               dim3(grid), dim3(256), 0, s, (const T*)gout,     \
+                               (const T*)x, (const T*)out, mask, save_mean,     \
+                               save_invstd, weight, k1, k2, (T*)gx, (T*)gres,   \
+                               nvec, C);                                        \
+        else                                                                    \
+            hipLaunchKernelGGL((bn_bwd_apply_kernel<T, VW, RELU_, false>), dim3(grid), \
                            dim3(256), 0, s, (const T*)gout, (const T*)x,        \
                            (const T*)out, mask, save_mean, save_invstd,        \
                            weight, k1, k2, (T*)gx, (T*)gres, nvec, C)

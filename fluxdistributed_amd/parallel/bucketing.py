@@ -124,6 +124,12 @@ class GradBucketer:
             b = self.buckets[self._next_launch]
             if self.world > 1:
                 seg = b["flat"][b["lo"]:b["hi"]]
+                if seg.is_cuda and dist.get_backend(self.pg) == "gloo":
+                    # gloo is stream-unaware: the host-side copy it makes
+                    # can read the segment before the in-flight grad
+                    # kernels land (RCCL orders against the current
+                    # stream; gloo needs an explicit sync)
+                    torch.cuda.current_stream(seg.device).synchronize()
                 self._works.append(
                     dist.all_reduce(seg, op=dist.ReduceOp.SUM,
                                     group=self.pg, async_op=True)
@@ -156,6 +162,9 @@ class GradBucketer:
         if self.world <= 1:
             return
         flats = {id(b["flat"]): b["flat"] for b in self.buckets}
+        if any(f.is_cuda for f in flats.values()) \
+                and dist.get_backend(self.pg) == "gloo":
+            torch.cuda.synchronize()
         works = [dist.all_reduce(f, op=dist.ReduceOp.SUM, group=self.pg, async_op=True)
                  for f in flats.values()]
         for w in works:
