@@ -156,7 +156,7 @@ class _BNAct(torch.autograd.Function):
         from .conv import take_conv_stats
 
         conv_part = take_conv_stats(x) if training else None
-        out, save_mean, save_invstd, relu_mask = C.bn_act_fwd(
+        out, save_mean, save_invstd = C.bn_act_fwd(
             x, weight, bias, running_mean, running_var,
             training, momentum, eps, relu,
             residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype),
@@ -164,8 +164,6 @@ class _BNAct(torch.autograd.Function):
         )
         ctx.save_for_backward(x, weight, save_mean, save_invstd, out)
         ctx.bn_bias = bias
-        ctx.relu_mask = (relu_mask if relu_mask is not None and
-                         relu_mask.numel() else None)
         ctx.relu = relu
         ctx.has_residual = residual is not None
         ctx.training = training
@@ -189,7 +187,6 @@ class _BNAct(torch.autograd.Function):
             grad_out.contiguous(memory_format=torch.channels_last),
             x, weight, save_mean, save_invstd, out, ctx.relu, ctx.training,
             gw_sl if direct else None, gb_sl if direct else None, want_gres,
-            ctx.relu_mask,
         )
         if direct:
             from ..parallel.bucketing import notify_grad_written
