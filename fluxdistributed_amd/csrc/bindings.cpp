@@ -71,11 +71,10 @@ at::Tensor add_relu_bwd(at::Tensor gout, at::Tensor out) {
     return gx;
 }
 
-std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
+std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
     at::Tensor x, at::Tensor weight, at::Tensor bias, at::Tensor running_mean,
     at::Tensor running_var, bool training, double momentum, double eps,
-    bool relu, at::Tensor residual, c10::optional<at::Tensor> conv_part,
-    c10::optional<at::Tensor> mask_out) {
+    bool relu, at::Tensor residual, c10::optional<at::Tensor> conv_part) {
     auto [rows, C] = nhwc_rows(x);
     const int V = dt_of(x) == DT::BF16 ? 8 : 4;
     TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
@@ -91,12 +90,6 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
     auto save_mean = at::empty({C}, fopts);
     auto save_invstd = at::empty({C}, fopts);
     auto out = at::empty_like(x);
-    // relu bitmask (one byte per V-vector): backward reads this instead of
-    // re-reading `out`. Caller-provided (arena slice) so steady-state steps
-    // never allocate.
-    at::Tensor mask;
-    if (training && relu && mask_out.has_value())
-        mask = *mask_out;
     auto stream = cur_stream();
 
     if (training && conv_part.has_value()) {
@@ -131,19 +124,16 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
                                 training, (float)momentum, (float)eps, stream);
     }
     fda::bn_apply_launch(x.data_ptr(), has_res ? resc.data_ptr() : nullptr,
-                         out.data_ptr(), ws.data_ptr<float>(),
-                         mask.defined() ? mask.data_ptr<uint8_t>() : nullptr,
-                         rows, (int)C, relu, dt_of(x), stream);
-    return {out, save_mean, save_invstd, mask};
+                         out.data_ptr(), ws.data_ptr<float>(), rows, (int)C,
+                         relu, dt_of(x), stream);
+    return {out, save_mean, save_invstd};
 }
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
     at::Tensor gout, at::Tensor x, at::Tensor weight, at::Tensor save_mean,
     at::Tensor save_invstd, at::Tensor out, bool relu, bool training,
     c10::optional<at::Tensor> gw_out, c10::optional<at::Tensor> gb_out,
-    bool want_gres, c10::optional<at::Tensor> relu_mask) {
-    const uint8_t* maskp =
-        relu_mask.has_value() ? relu_mask->data_ptr<uint8_t>() : nullptr;
+    bool want_gres) {
     // gw_out/gb_out set: fp32 flat-G slices, written += (direct grad)
     auto [rows, C] = nhwc_rows(x);
     auto gc = gout.contiguous(at::MemoryFormat::ChannelsLast);
@@ -159,7 +149,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
     auto part = at::empty({fda::bn_stats_partial_floats((int)C, rows, dt_of(x))},
                           fopts);
     fda::bn_bwd_stats_launch(gc.data_ptr(), x.data_ptr(), out.data_ptr(),
-                             maskp, save_mean.data_ptr<float>(),
+                             save_mean.data_ptr<float>(),
                              save_invstd.data_ptr<float>(), ws.data_ptr<float>(),
                              part.data_ptr<float>(), gw.data_ptr<float>(),
                              gb.data_ptr<float>(), rows, (int)C, relu,
@@ -167,7 +157,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_act_bwd(
     at::Tensor gres;
     if (want_gres) gres = at::empty_like(x);
     fda::bn_bwd_apply_launch(gc.data_ptr(), x.data_ptr(), out.data_ptr(),
-                             maskp, save_mean.data_ptr<float>(),
+                             save_mean.data_ptr<float>(),
                              save_invstd.data_ptr<float>(),
                              weight.data_ptr<float>(), ws.data_ptr<float>(),
                              gx.data_ptr(),
@@ -463,8 +453,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("training"), pybind11::arg("momentum"),
           pybind11::arg("eps"), pybind11::arg("relu"),
           pybind11::arg("residual"),
-          pybind11::arg("conv_part") = pybind11::none(),
-          pybind11::arg("mask_out") = pybind11::none());
+          pybind11::arg("conv_part") = pybind11::none());
     m.def("bn_act_bwd", &bn_act_bwd, pybind11::arg("gout"),
           pybind11::arg("x"), pybind11::arg("weight"),
           pybind11::arg("save_mean"), pybind11::arg("save_invstd"),
@@ -472,8 +461,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("training"),
           pybind11::arg("gw_out") = pybind11::none(),
           pybind11::arg("gb_out") = pybind11::none(),
-          pybind11::arg("want_gres") = false,
-          pybind11::arg("relu_mask") = pybind11::none());
+          pybind11::arg("want_gres") = false);
     m.def("maxpool_fwd", &maxpool_fwd);
     m.def("maxpool_bwd", &maxpool_bwd);
     m.def("gap_fwd", &gap_fwd);

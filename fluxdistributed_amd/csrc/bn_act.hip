@@ -189,11 +189,8 @@ template <typename T, int V, bool RELU, bool RES>
 __global__ void bn_apply_kernel(const T* __restrict__ x,
                                 const T* __restrict__ res, T* __restrict__ out,
                                 const float* __restrict__ scale,
-                                const float* __restrict__ shift,
-                                unsigned char* __restrict__ mask,  // nullable
-                                int64_t nvec, int C) {
-    // mask: one byte per V-vector, bit k = (out element k > 0) — backward
-    // reads 1/16th the bytes instead of re-reading `out` for the mask.
+                                const float* __restrict__ shift, int64_t nvec,
+                                int C) {
     const int cvec = C / V;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
          i += (int64_t)gridDim.x * blockDim.x) {
@@ -201,33 +198,23 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
         T xv[V], rv_[V], ov[V];
         *(uint4*)xv = ((const uint4*)x)[i];
         if constexpr (RES) *(uint4*)rv_ = ((const uint4*)res)[i];
-        unsigned m = 0;
         #pragma unroll
         for (int k = 0; k < V; ++k) {
             float v = load_f32(xv + k) * scale[c0 + k] + shift[c0 + k];
             if constexpr (RES) v += load_f32(rv_ + k);
-            if constexpr (RELU) {
-                m |= (v > 0.f ? 1u : 0u) << k;
-                v = fmaxf(v, 0.f);
-            }
+            if constexpr (RELU) v = fmaxf(v, 0.f);
             store_f32(ov + k, v);
         }
         ((uint4*)out)[i] = *(uint4*)ov;
-        if (RELU && mask != nullptr) mask[i] = (unsigned char)m;
     }
 }
 
 // ---- backward -------------------------------------------------------------
 
-template <typename T, int V, bool RELU, bool MASKED>
+template <typename T, int V, bool RELU>
 __global__ void bn_bwd_stats_kernel(const T* __restrict__ gout,
                                     const T* __restrict__ x,
                                     const T* __restrict__ out,
-                                    const unsigned char* __restrict__ mask,
-                                    // MASKED is compile-time: a runtime
-                                    // mask!=null branch around the loads
-                                    // de-pipelines the streaming loop
-                                    // (guide §5 trap (c))
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
                                     float* __restrict__ part, int64_t rows,
@@ -251,18 +238,11 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ gout,
         T gv[V], xv[V], ov[V];
         *(uint4*)gv = *(const uint4*)(gout + r * C + c0);
         *(uint4*)xv = *(const uint4*)(x + r * C + c0);
-        unsigned mbits = 0xffu;
-        if constexpr (RELU && MASKED)
-            mbits = mask[(r * C + c0) / V];
-        else if constexpr (RELU)
-            *(uint4*)ov = *(const uint4*)(out + r * C + c0);
+        if constexpr (RELU) *(uint4*)ov = *(const uint4*)(out + r * C + c0);
         #pragma unroll
         for (int k = 0; k < V; ++k) {
             float g = load_f32(gv + k);
-            if constexpr (RELU && MASKED)
-                g = ((mbits >> k) & 1u) ? g : 0.f;
-            else if constexpr (RELU)
-                g = load_f32(ov + k) > 0.f ? g : 0.f;
+            if constexpr (RELU) g = load_f32(ov + k) > 0.f ? g : 0.f;
             const float xhat = (load_f32(xv + k) - mu[k]) * is[k];
             sg[k] += g;
             sgx[k] += g * xhat;
@@ -292,11 +272,10 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ gout,
     }
 }
 
-template <typename T, int V, bool RELU, bool MASKED>
+template <typename T, int V, bool RELU>
 __global__ void bn_bwd_apply_kernel(const T* __restrict__ gout,
                                     const T* __restrict__ x,
                                     const T* __restrict__ out,
-                                    const unsigned char* __restrict__ mask,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
                                     const float* __restrict__ weight,
@@ -315,19 +294,12 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ gout,
         T gv[V], xv[V], ov[V], rv_[V], mv[V];
         *(uint4*)gv = ((const uint4*)gout)[i];
         *(uint4*)xv = ((const uint4*)x)[i];
-        unsigned mbits = 0xffu;
-        if constexpr (RELU && MASKED)
-            mbits = mask[i];
-        else if constexpr (RELU)
-            *(uint4*)ov = ((const uint4*)out)[i];
+        if constexpr (RELU) *(uint4*)ov = ((const uint4*)out)[i];
         #pragma unroll
         for (int k = 0; k < V; ++k) {
             const int c = c0 + k;
             float g = load_f32(gv + k);
-            if constexpr (RELU && MASKED)
-                g = ((mbits >> k) & 1u) ? g : 0.f;
-            else if constexpr (RELU)
-                g = load_f32(ov + k) > 0.f ? g : 0.f;
+            if constexpr (RELU) g = load_f32(ov + k) > 0.f ? g : 0.f;
             store_f32(mv + k, g);
             const float is = invstd[c];
             const float xhat = (load_f32(xv + k) - mean[c]) * is;
@@ -399,8 +371,8 @@ void bn_finalize_launch(float* ws, const float* weight, const float* bias,
 }
 
 void bn_apply_launch(const void* x, const void* residual, void* out,
-                     const float* ws, unsigned char* mask, int64_t rows,
-                     int C, bool relu, DT dt, hipStream_t s) {
+                     const float* ws, int64_t rows, int C, bool relu, DT dt,
+                     hipStream_t s) {
     const int V = dt == DT::BF16 ? 8 : 4;
     const int64_t nvec = rows * C / V;
     int64_t blocks = (nvec + 255) / 256;
@@ -410,7 +382,7 @@ void bn_apply_launch(const void* x, const void* residual, void* out,
     #define FDA_APPLY(T, VW, RELU_, RES_)                                       \
         hipLaunchKernelGGL((bn_apply_kernel<T, VW, RELU_, RES_>), dim3(grid),   \
                            dim3(256), 0, s, (const T*)x, (const T*)residual,    \
-                           (T*)out, scale, shift, mask, nvec, C)
+                           (T*)out, scale, shift, nvec, C)
     if (dt == DT::BF16) {
         if (relu) { if (residual) FDA_APPLY(unsigned short, 8, true, true); else FDA_APPLY(unsigned short, 8, true, false); }
         else      { if (residual) FDA_APPLY(unsigned short, 8, false, true); else FDA_APPLY(unsigned short, 8, false, false); }
@@ -422,7 +394,6 @@ void bn_apply_launch(const void* x, const void* residual, void* out,
 }
 
 void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
-                         const unsigned char* mask,
                          const float* save_mean, const float* save_invstd,
                          float* ws, float* part, float* gw, float* gb,
                          int64_t rows, int C, bool relu, bool training,
@@ -434,16 +405,9 @@ void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
         const int c_base = ch * chunkC;
         const int cc = (C - c_base) < chunkC ? (C - c_base) : chunkC;
         #define FDA_BSTATS(T, VW, RELU_)                                        \
-            if (mask != nullptr)                                                \
-                hipLaunchKernelGGL((bn_bwd_stats_kernel<T, VW, RELU_, true>),   \
-                                   dim3(grid), dim3(256), shmem, s,             \
-                                   (const T*)gout, (const T*)x, (const T*)out,  \
-                                   mask, save_mean, save_invstd, part, rows, C, \
-                                   c_base, cc);                                 \
-            else                                                                \
-                hipLaunchKernelGGL((bn_bwd_stats_kernel<T, VW, RELU_, false>), dim3(grid), \
+            hipLaunchKernelGGL((bn_bwd_stats_kernel<T, VW, RELU_>), dim3(grid), \
                                dim3(256), shmem, s, (const T*)gout,             \
-                               (const T*)x, (const T*)out, mask, save_mean,     \
+                               (const T*)x, (const T*)out, save_mean,           \
                                save_invstd, part, rows, C, c_base, cc)
         if (dt == DT::BF16) { if (relu) FDA_BSTATS(unsigned short, 8, true); else FDA_BSTATS(unsigned short, 8, false); }
         else { if (relu) FDA_BSTATS(float, 4, true); else FDA_BSTATS(float, 4, false); }
@@ -468,7 +432,6 @@ void bn_finalize_from_partials_launch(
 }
 
 void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
-                         const unsigned char* mask,
                          const float* save_mean, const float* save_invstd,
                          const float* weight, const float* ws, void* gx,
                          void* gres, int64_t rows, int C, bool relu,
@@ -480,17 +443,10 @@ void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
     const float* k1 = ws + 2 * C;
     const float* k2 = ws + 3 * C;
     #define FDA_BAPPLY(T, VW, RELU_)                                            \
-        if (mask != nullptr)                                                    \
-            hipLaunchKernelGGL((bn_bwd_apply_kernel<T, VW, RELU_, true>),       \
-                               dim3(grid), dim3(256), 0, s, (const T*)gout,     \
-                               (const T*)x, (const T*)out, mask, save_mean,     \
-                               save_invstd, weight, k1, k2, (T*)gx, (T*)gres,   \
-                               nvec, C);                                        \
-        else                                                                    \
-            hipLaunchKernelGGL((bn_bwd_apply_kernel<T, VW, RELU_, false>), dim3(grid), \
+        hipLaunchKernelGGL((bn_bwd_apply_kernel<T, VW, RELU_>), dim3(grid),     \
                            dim3(256), 0, s, (const T*)gout, (const T*)x,        \
-                           (const T*)out, mask, save_mean, save_invstd,        \
-                           weight, k1, k2, (T*)gx, (T*)gres, nvec, C)
+                           (const T*)out, save_mean, save_invstd, weight, k1,   \
+                           k2, (T*)gx, (T*)gres, nvec, C)
     if (dt == DT::BF16) { if (relu) FDA_BAPPLY(unsigned short, 8, true); else FDA_BAPPLY(unsigned short, 8, false); }
     else { if (relu) FDA_BAPPLY(float, 4, true); else FDA_BAPPLY(float, 4, false); }
     #undef FDA_BAPPLY

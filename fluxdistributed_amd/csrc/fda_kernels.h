@@ -44,20 +44,18 @@ void bn_finalize_launch(float* ws, const float* weight, const float* bias,
                         int C, bool training, float momentum, float eps,
                         hipStream_t s);
 void bn_apply_launch(const void* x, const void* residual, void* out,
-                     const float* ws /*scale/shift*/, unsigned char* mask,
-                     int64_t rows, int C, bool relu, DT dt, hipStream_t s);
+                     const float* ws /*scale/shift*/, int64_t rows, int C,
+                     bool relu, DT dt, hipStream_t s);
 
 // backward. ws layout (floats): [0,C) sum_g  [C,2C) sum_g_xhat
 //   [2C,3C) k1  [3C,4C) k2   (k's folded with invstd*gamma in finalize)
 // bwd: stats partials + fused reduce/finalize (gw, gb, k1/k2 into ws)
 void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
-                         const unsigned char* mask,
                          const float* save_mean, const float* save_invstd,
                          float* ws, float* part, float* gw, float* gb,
                          int64_t rows, int C, bool relu, bool training,
                          bool accum_g, DT dt, hipStream_t s);
 void bn_bwd_apply_launch(const void* gout, const void* x, const void* out,
-                         const unsigned char* mask,
                          const float* save_mean, const float* save_invstd,
                          const float* weight, const float* ws, void* gx,
                          void* gres, int64_t rows, int C, bool relu,

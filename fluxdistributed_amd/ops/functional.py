@@ -137,40 +137,6 @@ class MaxPool2d(torch.nn.Module):
 # --------------------------------------------------------------------------
 
 
-class _MaskArena:
-    """Per-device byte arena for BN relu masks, keyed by the BN weight
-    parameter — steady-state steps never allocate (a per-layer mask alloc
-    measured as an allocator-churn regression)."""
-
-    def __init__(self, device):
-        self.device = device
-        self.slices = {}
-        self.sizes = []
-        self.arena = None
-
-    def get(self, key, nbytes):
-        if key not in self.slices:
-            self.slices[key] = (sum(self.sizes), nbytes)
-            self.sizes.append(nbytes)
-            self.arena = None
-        if self.arena is None:
-            self.arena = torch.empty(sum(self.sizes), dtype=torch.uint8,
-                                     device=self.device)
-        off, n = self.slices[key]
-        assert n == nbytes
-        return self.arena[off : off + n]
-
-
-_MASK_ARENAS: dict = {}
-
-
-def _mask_for(weight, nbytes):
-    dev = weight.device
-    if dev not in _MASK_ARENAS:
-        _MASK_ARENAS[dev] = _MaskArena(dev)
-    return _MASK_ARENAS[dev].get(id(weight), nbytes)
-
-
 class _BNAct(torch.autograd.Function):
     @staticmethod
     def forward(
@@ -190,20 +156,14 @@ class _BNAct(torch.autograd.Function):
         from .conv import take_conv_stats
 
         conv_part = take_conv_stats(x) if training else None
-        mask_out = None
-        if training and relu and torch.is_grad_enabled():
-            V = 8 if x.dtype == torch.bfloat16 else 4
-            mask_out = _mask_for(weight, x.numel() // V)
-        out, save_mean, save_invstd, relu_mask = C.bn_act_fwd(
+        out, save_mean, save_invstd = C.bn_act_fwd(
             x, weight, bias, running_mean, running_var,
             training, momentum, eps, relu,
             residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype),
-            conv_part, mask_out,
+            conv_part,
         )
         ctx.save_for_backward(x, weight, save_mean, save_invstd, out)
         ctx.bn_bias = bias
-        ctx.relu_mask = (relu_mask if relu_mask is not None and
-                         relu_mask.numel() else None)
         ctx.relu = relu
         ctx.has_residual = residual is not None
         ctx.training = training
@@ -227,7 +187,6 @@ class _BNAct(torch.autograd.Function):
             grad_out.contiguous(memory_format=torch.channels_last),
             x, weight, save_mean, save_invstd, out, ctx.relu, ctx.training,
             gw_sl if direct else None, gb_sl if direct else None, want_gres,
-            ctx.relu_mask,
         )
         if direct:
             from ..parallel.bucketing import notify_grad_written

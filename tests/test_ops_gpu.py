@@ -67,7 +67,7 @@ def test_bn_act_fwd(dtype, tol, Cc, training, relu, res, seed):
     residual = (torch.randn_like(x).contiguous(memory_format=torch.channels_last)
                 if res else torch.empty(0, device=DEV, dtype=dtype))
 
-    out, mean, invstd, _m = C.bn_act_fwd(x, w, b, rm, rv, training, 0.1, 1e-5, relu,
+    out, mean, invstd = C.bn_act_fwd(x, w, b, rm, rv, training, 0.1, 1e-5, relu,
                                      residual)
 
     xf = x.float()
@@ -93,7 +93,7 @@ def test_bn_act_bwd(dtype, tol, Cc, relu, seed):
     w = (torch.rand(Cc, device=DEV) + 0.5)
     b = torch.randn(Cc, device=DEV)
     rm, rv = torch.zeros(Cc, device=DEV), torch.ones(Cc, device=DEV)
-    out, mean, invstd, _m = C.bn_act_fwd(x, w, b, rm, rv, True, 0.1, 1e-5, relu,
+    out, mean, invstd = C.bn_act_fwd(x, w, b, rm, rv, True, 0.1, 1e-5, relu,
                                      torch.empty(0, device=DEV, dtype=dtype))
     gout = torch.randn_like(x).contiguous(memory_format=torch.channels_last)
     gx, gw, gb, _ = C.bn_act_bwd(gout, x, w, mean, invstd, out, relu, True)
