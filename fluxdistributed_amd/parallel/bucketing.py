@@ -14,6 +14,7 @@ fused optimizer's G), and buckets are contiguous slices of it — zero
 packing copies.
 """
 
+import threading
 from typing import List, Optional
 
 import torch
@@ -56,6 +57,7 @@ class GradBucketer:
         self._build(optimizer, bucket_cap_mb)
         self._works: List = []
         self._next_launch = 0
+        self._lock = threading.Lock()   # notify may fire from engine threads
 
     def _build(self, optimizer: _FlatOptimizer, cap_mb: float):
         cap = int(cap_mb * 1e6)
@@ -120,6 +122,10 @@ class GradBucketer:
             self._launch_ready()
 
     def _launch_ready(self):
+        with self._lock:
+            self._launch_ready_locked()
+
+    def _launch_ready_locked(self):
         while self._next_launch < len(self.buckets) and self._ready[self._next_launch]:
             b = self.buckets[self._next_launch]
             if self.world > 1:
