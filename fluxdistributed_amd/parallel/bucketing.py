@@ -131,15 +131,18 @@ class GradBucketer:
             if self.world > 1:
                 seg = b["flat"][b["lo"]:b["hi"]]
                 if seg.is_cuda and dist.get_backend(self.pg) == "gloo":
-                    # gloo is stream-unaware: the host-side copy it makes
-                    # can read the segment before the in-flight grad
-                    # kernels land (RCCL orders against the current
-                    # stream; gloo needs an explicit sync)
+                    # gloo's async CUDA collectives stage through host
+                    # buffers that race when several are in flight
+                    # (replica divergence measured in the 2-rank GPU
+                    # test); run them synchronously — gloo is only the
+                    # fallback transport, RCCL below keeps true overlap.
                     torch.cuda.current_stream(seg.device).synchronize()
-                self._works.append(
-                    dist.all_reduce(seg, op=dist.ReduceOp.SUM,
-                                    group=self.pg, async_op=True)
-                )
+                    dist.all_reduce(seg, op=dist.ReduceOp.SUM, group=self.pg)
+                else:
+                    self._works.append(
+                        dist.all_reduce(seg, op=dist.ReduceOp.SUM,
+                                        group=self.pg, async_op=True)
+                    )
             self._next_launch += 1
 
     def finalize(self):
