@@ -72,6 +72,7 @@ class GradBucketer:
                     self.buckets.append(dict(
                         flat=g.G, lo=cur_lo, hi=cur_hi,
                         params=set(cur_params), pending=len(cur_params),
+                        seen=set(),
                     ))
                 cur_params, cur_lo, cur_hi, cur_bytes = [], None, None, 0
             for idx in order:
@@ -116,6 +117,13 @@ class GradBucketer:
     def _on_grad(self, p):
         bi = self._param2bucket[id(p)]
         b = self.buckets[bi]
+        # Idempotent per step: a direct-grad param fires BOTH our notify AND
+        # the post-accumulate hook (the engine still runs AccumulateGrad for
+        # a None grad). Double-counting launched buckets early and the late
+        # gradients were never all-reduced (caught by the 2-rank GPU test).
+        if id(p) in b["seen"]:
+            return
+        b["seen"].add(id(p))
         b["pending"] -= 1
         if b["pending"] == 0:
             self._ready[bi] = True
@@ -162,6 +170,7 @@ class GradBucketer:
                 b["flat"][b["lo"]:b["hi"]].div_(self.world)
         for bi, b in enumerate(self.buckets):
             b["pending"] = len(b["params"])
+            b["seen"] = set()
             self._ready[bi] = False
         self._next_launch = 0
 
@@ -183,5 +192,6 @@ class GradBucketer:
                 f.div_(self.world)
         for bi, b in enumerate(self.buckets):
             b["pending"] = len(b["params"])
+            b["seen"] = set()
             self._ready[bi] = False
         self._next_launch = 0
