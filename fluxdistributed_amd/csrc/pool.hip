@@ -101,9 +101,16 @@ __global__ void maxpool_bwd_kernel(const T* __restrict__ gout,
                 const int64_t o = (((int64_t)n * HO + ho) * WO + wo) * C + c0;
                 T gv[V];
                 *(uint4*)gv = *(const uint4*)(gout + o);
+                // V idx bytes in one word (per-byte loads were issue-bound)
+                uint64_t ib;
+                if constexpr (V == 8)
+                    ib = *(const uint64_t*)(idx + o);
+                else
+                    ib = *(const uint32_t*)(idx + o);
                 #pragma unroll
                 for (int k = 0; k < V; ++k)
-                    if (idx[o + k] == p) acc[k] += load_f32(gv + k);
+                    if (((ib >> (8 * k)) & 0xffu) == p)
+                        acc[k] += load_f32(gv + k);
             }
         }
         T rv[V];
