@@ -15,7 +15,7 @@ The DDP all-reduce can be captured too (RCCL supports hipGraph capture);
 overlapped bucketed comm use the eager step (bucket hooks are host-side).
 """
 
-from typing import Callable, Optional
+from typing import Callable
 
 import torch
 

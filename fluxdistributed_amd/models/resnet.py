@@ -14,7 +14,7 @@ from typing import List, Optional, Type
 import torch
 import torch.nn as nn
 
-from ..ops.functional import (batch_norm_act, fused_add_relu, MaxPool2d,
+from ..ops.functional import (batch_norm_act, MaxPool2d,
                               GlobalAvgPool)
 
 

@@ -15,8 +15,8 @@ One kernel launch updates the whole model (~22M elems for ResNet-34): pure
 HBM-bound streaming, vectorized 16B/lane (guide Appendix B elementwise).
 """
 
-import math
-from typing import Iterable, List
+
+from typing import List
 
 import torch
 

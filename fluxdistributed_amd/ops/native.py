@@ -6,7 +6,7 @@ built in-tree so the .so travels with the repo snapshot to GPU machines.
 
 import os
 
-import torch
+
 
 _NATIVE = None
 _TRIED = False

@@ -15,7 +15,7 @@ packing copies.
 """
 
 import threading
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.distributed as dist
