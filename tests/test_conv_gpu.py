@@ -26,6 +26,10 @@ SHAPES = [
     (4, 256, 14, 14, 512, 1, 2),
     (96, 512, 7, 7, 512, 3, 1),    # M=4704 not a multiple of 128
     (2, 64, 9, 11, 64, 3, 1),      # odd spatial
+    (3, 64, 113, 113, 128, 3, 2),  # odd spatial + stride 2 (parity classes)
+    (2, 128, 14, 14, 512, 1, 1),   # bottleneck expand 1x1
+    (2, 256, 27, 27, 64, 1, 2),    # odd spatial 1x1 s2
+    (2, 64, 56, 56, 256, 1, 1),    # bottleneck downsample-free expand
 ]
 
 
