@@ -47,6 +47,11 @@ def main():
     model.eval()
     device = torch.device("cuda:0") if torch.cuda.is_available() else torch.device("cpu")
     model = model.to(device)
+    if device.type == "cuda":
+        # inference on the native kernel path: channels_last bf16
+        from fluxdistributed_amd.utils.precision import to_mixed_bf16
+
+        model = to_mixed_bf16(model.to(memory_format=torch.channels_last))
 
     names = None
     if args.labels:
@@ -55,6 +60,8 @@ def main():
         names = [desc for _, desc in labels(args.labels)]
 
     x = torch.stack([load_image(pth) for pth in args.images]).to(device)
+    if device.type == "cuda":
+        x = x.bfloat16().contiguous(memory_format=torch.channels_last)
     with torch.no_grad():
         logits = model(x).float().cpu()
     probs = torch.softmax(logits, dim=-1)

@@ -156,6 +156,7 @@ class _BNAct(torch.autograd.Function):
         from .conv import take_conv_stats
 
         conv_part = take_conv_stats(x) if training else None
+        x = x.contiguous(memory_format=torch.channels_last)
         out, save_mean, save_invstd = C.bn_act_fwd(
             x, weight, bias, running_mean, running_var,
             training, momentum, eps, relu,
