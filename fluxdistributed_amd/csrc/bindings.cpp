@@ -232,6 +232,16 @@ void adam_step(at::Tensor P, at::Tensor G, at::Tensor M, at::Tensor V,
 
 // ---- implicit-GEMM conv (NHWC bf16, conv_igemm.hip) -----------------------
 
+// split-K when the launch would underfill the 256-CU chip (layer4-size
+// shapes run at ~30% fill otherwise)
+static int conv_pick_sk(long M, int OC, int zbase) {
+    const int BM = (OC % 128 == 0) ? 128 : 256;
+    const int BN = (OC % 128 == 0) ? 128 : 64;
+    const long blocks = ((M + BM - 1) / BM) * (OC / BN) * zbase;
+    if (blocks >= 384) return 1;
+    return blocks >= 192 ? 2 : 4;
+}
+
 at::Tensor conv_igemm_fwd(at::Tensor x, at::Tensor w,
                           int64_t sy, int64_t sx, int64_t py, int64_t px) {
     TORCH_CHECK(x.dim() == 4 && w.dim() == 4);
@@ -251,6 +261,21 @@ at::Tensor conv_igemm_fwd(at::Tensor x, at::Tensor w,
     const int Q = (W + 2 * (int)px - S) / (int)sx + 1;
     auto y = at::empty({N, K, P, Q},
                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    const long M = (long)N * P * Q;
+    const int SK = conv_pick_sk(M, K, /*zbase=*/1);
+    if (SK > 1) {
+        auto part = at::empty({(long)SK * M * K},
+                              x.options().dtype(at::kFloat));
+        fda::conv_igemm_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                               N, H, W, C, K, P, Q, R, S, (int)sy, (int)sx,
+                               (int)py, (int)px, false, cur_stream(), nullptr,
+                               part.data_ptr<float>(), SK);
+        fda::conv_skcombine_launch(part.data_ptr<float>(), y.data_ptr(),
+                                   nullptr, M, K, SK,
+                                   fda::conv_skcombine_blocks(M, K),
+                                   cur_stream());
+        return y;
+    }
     fda::conv_igemm_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
                            N, H, W, C, K, P, Q, R, S, (int)sy, (int)sx,
                            (int)py, (int)px, /*dgrad=*/false, cur_stream());
@@ -273,10 +298,24 @@ std::tuple<at::Tensor, at::Tensor> conv_igemm_fwd_stats(
     const int P = (H + 2 * (int)py - R) / (int)sy + 1;
     const int Q = (W + 2 * (int)px - S) / (int)sx + 1;
     const long M = (long)N * P * Q;
-    const int BM = (K % 128 == 0) ? 128 : 256;
-    const long mtiles = (M + BM - 1) / BM;
     auto y = at::empty({N, K, P, Q},
                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    const int SK = conv_pick_sk(M, K, 1);
+    if (SK > 1) {
+        const int nb = fda::conv_skcombine_blocks(M, K);
+        auto skp = at::empty({(long)SK * M * K}, x.options().dtype(at::kFloat));
+        auto part = at::empty({nb, 2, K}, x.options().dtype(at::kFloat));
+        fda::conv_igemm_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                               N, H, W, C, K, P, Q, R, S, (int)sy, (int)sx,
+                               (int)py, (int)px, false, cur_stream(), nullptr,
+                               skp.data_ptr<float>(), SK);
+        fda::conv_skcombine_launch(skp.data_ptr<float>(), y.data_ptr(),
+                                   part.data_ptr<float>(), M, K, SK, nb,
+                                   cur_stream());
+        return {y, part};
+    }
+    const int BM = (K % 128 == 0) ? 128 : 256;
+    const long mtiles = (M + BM - 1) / BM;
     auto part = at::empty({mtiles, 2, K}, x.options().dtype(at::kFloat));
     fda::conv_igemm_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
                            N, H, W, C, K, P, Q, R, S, (int)sy, (int)sx,
@@ -302,6 +341,23 @@ at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
                 "conv_igemm: C and K must be multiples of 64");
     auto dx = at::empty({N, C, H, W},
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+    const long M = (long)N * H * W;
+    const int SK = (sy == 1 && sx == 1)
+        ? conv_pick_sk(M, (int)C, 1) : 1;   // parity classes already fan out
+    if (SK > 1) {
+        auto skp = at::empty({(long)SK * M * C},
+                             dy.options().dtype(at::kFloat));
+        fda::conv_igemm_launch(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
+                               N, (int)H, (int)W, (int)C, K, P, Q, (int)R,
+                               (int)S, (int)sy, (int)sx, (int)py, (int)px,
+                               true, cur_stream(), nullptr,
+                               skp.data_ptr<float>(), SK);
+        fda::conv_skcombine_launch(skp.data_ptr<float>(), dx.data_ptr(),
+                                   nullptr, M, (int)C, SK,
+                                   fda::conv_skcombine_blocks(M, (int)C),
+                                   cur_stream());
+        return dx;
+    }
     fda::conv_igemm_launch(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                            N, (int)H, (int)W, (int)C, K, P, Q, (int)R, (int)S,
                            (int)sy, (int)sx, (int)py, (int)px, /*dgrad=*/true,

@@ -65,10 +65,15 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     int N, int H, int W, int C,
     int K, int P, int Q,
     int R, int S, int sy, int sx, int py, int px,
-    float* __restrict__ stats /* [mtiles][2][OC] or null: per-channel
+    float* __restrict__ stats, /* [mtiles][2][OC] or null: per-channel
                                  sum/sumsq of the rounded output — feeds the
                                  BN reduce+finalize directly (the separate
-                                 bn_stats read pass is skipped) */) {
+                                 bn_stats read pass is skipped) */
+    float* __restrict__ skpart, /* split-K: fp32 partial output
+                                 [SK][M][OC]; combine kernel reduces+casts.
+                                 Small-M late layers fill only ~30% of the
+                                 chip otherwise. */
+    int SK) {
     constexpr int A_ELEMS = BM * BK;
     constexpr int B_ELEMS = BN * BK;
     constexpr int BUF_ELEMS = A_ELEMS + B_ELEMS;
@@ -79,11 +84,13 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     const int RC = (MODE == CONV_FWD) ? C : ((MODE == CONV_STEM) ? 64 : K);
 
     int a = 0, b = 0, OH, OW, r0 = 0, s0 = 0, nR = R, nS = S;
+    int sk = 0, zrest = blockIdx.z;
+    if (SK > 1) { sk = zrest % SK; zrest /= SK; }
     if (MODE != CONV_DGRAD) {
         OH = P; OW = Q;
         if (MODE == CONV_STEM) nS = 1;   // taps iterate r only
     } else {
-        a = blockIdx.z / sx;  b = blockIdx.z % sx;
+        a = zrest / sx;  b = zrest % sx;
         OH = (H - a + sy - 1) / sy;
         OW = (W - b + sx - 1) / sx;
         r0 = (a + py) % sy;  nR = (R - r0 + sy - 1) / sy;
@@ -240,15 +247,22 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
 
     // ---- main loop: double buffer; tile t+1's DMA in flight over tile t's
     // compute, drained at the iteration boundary (guide T3 minimum form) ---
-    if (T > 0) stage(0, 0);
-    if (NBUF > 2 && T > 1) stage(1, 1);
-    for (int it = 0; it < T; ++it) {
+    int it0 = 0, itN = T;
+    if (SK > 1) {
+        const int chunk = (T + SK - 1) / SK;
+        it0 = sk * chunk;
+        itN = it0 + chunk < T ? it0 + chunk : T;
+        if (it0 >= itN) itN = it0;       // empty slice: epilogue writes 0
+    }
+    if (it0 < itN) stage(it0 % NBUF, it0);
+    if (NBUF > 2 && it0 + 1 < itN) stage((it0 + 1) % NBUF, it0 + 1);
+    for (int it = it0; it < itN; ++it) {
         // tile `it` landed chip-wide: each wave drains its own DMA, the
         // barrier joins all waves. This is the loop's ONLY barrier — the
         // next K-step's staging targets the buffer every wave finished
         // reading before it arrived here. With NBUF==3 the counted wait
         // leaves tile it+1's DMA in flight across the barrier (T4).
-        if (NBUF > 2 && it + 1 < T) {
+        if (NBUF > 2 && it + 1 < itN) {
             if (GPW == 8)
                 asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
             else
@@ -257,7 +271,7 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
         __builtin_amdgcn_s_barrier();
-        if (it + NBUF - 1 < T) stage((it + NBUF - 1) % NBUF, it + NBUF - 1);
+        if (it + NBUF - 1 < itN) stage((it + NBUF - 1) % NBUF, it + NBUF - 1);
         const unsigned short* buf = lds + (it % NBUF) * BUF_ELEMS;
         short8 af[4][2], bf[4][2];
         #pragma unroll
@@ -305,6 +319,15 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
                 const int n = (int)(m / ((long)OW * OH));
                 obase = (((long)n * H + a + (long)sy * hh) * W + b +
                          (long)sx * ww) * C;
+            }
+            if (skpart != nullptr) {
+                // split-K: fp32 partials, linear by m (the dgrad split path
+                // is stride-1 only, where obase == m*OC)
+                float* prow = skpart + ((long)sk * M + m) * OC + n0 + wn * 64;
+                #pragma unroll
+                for (int ni = 0; ni < 4; ++ni)
+                    prow[ni * 16 + fcol] = acc[mi][ni][j];
+                continue;
             }
             unsigned short* orow = out + obase + n0 + wn * 64;
             #pragma unroll
@@ -360,13 +383,15 @@ template <int MODE, int BM, int BN, int WN, int NBUF = 2>
 static void launch_cfg(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
-                       hipStream_t stream, float* stats = nullptr) {
+                       hipStream_t stream, float* stats = nullptr,
+                       float* skpart = nullptr, int SK = 1) {
     const int OC = (MODE == CONV_DGRAD) ? C : K;
     const long M = (MODE != CONV_DGRAD)
         ? (long)N * P * Q
         : (long)N * ((H + sy - 1) / sy) * ((W + sx - 1) / sx);
+    const unsigned zbase = (MODE != CONV_DGRAD) ? 1u : (unsigned)(sy * sx);
     dim3 grid((unsigned)((M + BM - 1) / BM), (unsigned)(OC / BN),
-              (MODE != CONV_DGRAD) ? 1u : (unsigned)(sy * sx));
+              zbase * (unsigned)SK);
     const size_t shmem = NBUF * (BM * BK + BN * BK) * sizeof(unsigned short);
     if (shmem > 65536) {
         static bool raised = [] {
@@ -380,7 +405,84 @@ static void launch_cfg(const void* src, const void* wgt, void* out,
     hipLaunchKernelGGL((conv_igemm_kernel<MODE, BM, BN, WN, NBUF>), grid, dim3(256),
                        shmem, stream, (const unsigned short*)src,
                        (const unsigned short*)wgt, (unsigned short*)out,
-                       N, H, W, C, K, P, Q, R, S, sy, sx, py, px, stats);
+                       N, H, W, C, K, P, Q, R, S, sy, sx, py, px, stats,
+                       skpart, SK);
+}
+
+// ---- split-K combine: y = bf16(sum_sk part) (+ BN stats partials) --------
+__global__ __launch_bounds__(256) void conv_skcombine_kernel(
+    const float* __restrict__ part, unsigned short* __restrict__ y,
+    float* __restrict__ stats, long M, int OC, int SK) {
+    __shared__ float smem[2048];          // 256 threads x 8 lanes
+    const int V = 8;
+    const int tpr = OC / V;               // threads per row (OC <= 2048)
+    const int rpb = 256 / tpr;
+    const int lane_c = threadIdx.x % tpr;
+    const int sub_r = threadIdx.x / tpr;
+    const int c0 = lane_c * V;
+    float s[V], q[V];
+    #pragma unroll
+    for (int k = 0; k < V; ++k) s[k] = q[k] = 0.f;
+    for (long r = (long)blockIdx.x * rpb + sub_r; r < M;
+         r += (long)gridDim.x * rpb) {
+        float v[V];
+        #pragma unroll
+        for (int k = 0; k < V; ++k) v[k] = 0.f;
+        for (int z = 0; z < SK; ++z) {
+            const float* p = part + ((long)z * M + r) * OC + c0;
+            float4 a = *(const float4*)p;
+            float4 b = *(const float4*)(p + 4);
+            v[0] += a.x; v[1] += a.y; v[2] += a.z; v[3] += a.w;
+            v[4] += b.x; v[5] += b.y; v[6] += b.z; v[7] += b.w;
+        }
+        unsigned short o[V];
+        #pragma unroll
+        for (int k = 0; k < V; ++k) {
+            o[k] = f32_to_bf16bits(v[k]);
+            if (stats != nullptr) {
+                const float vr = bf16bits_to_f32(o[k]);
+                s[k] += vr;
+                q[k] += vr * vr;
+            }
+        }
+        *(uint4*)(y + r * OC + c0) = *(uint4*)o;
+    }
+    if (stats == nullptr) return;
+    // fold the rpb row-groups, write this block's [2][OC] partial slice
+    float* outp = stats + (long)blockIdx.x * 2 * OC;
+    #pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+        float* loc = pass == 0 ? s : q;
+        #pragma unroll
+        for (int k = 0; k < V; ++k) smem[threadIdx.x * V + k] = loc[k];
+        __syncthreads();
+        if (sub_r == 0) {
+            float acc[V];
+            #pragma unroll
+            for (int k = 0; k < V; ++k) acc[k] = 0.f;
+            for (int rr = 0; rr < rpb; ++rr) {
+                const float* sp = smem + (rr * tpr + lane_c) * V;
+                #pragma unroll
+                for (int k = 0; k < V; ++k) acc[k] += sp[k];
+            }
+            #pragma unroll
+            for (int k = 0; k < V; ++k)
+                outp[pass * OC + lane_c * V + k] = acc[k];
+        }
+        __syncthreads();
+    }
+}
+
+int conv_skcombine_blocks(long M, int OC) {
+    const int rpb = 256 / (OC / 8);
+    long b = (M + rpb - 1) / rpb;
+    return (int)(b < 512 ? b : 512);
+}
+
+void conv_skcombine_launch(const float* part, void* y, float* stats, long M,
+                           int OC, int SK, int nblocks, hipStream_t stream) {
+    hipLaunchKernelGGL(conv_skcombine_kernel, dim3(nblocks), dim3(256), 0,
+                       stream, part, (unsigned short*)y, stats, M, OC, SK);
 }
 
 void conv_stem_fwd_launch(const void* src, const void* wgt, void* out,
@@ -407,7 +509,8 @@ static int conv_nbuf() {
 void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
-                       bool dgrad, hipStream_t stream, float* stats) {
+                       bool dgrad, hipStream_t stream, float* stats,
+                       float* skpart, int SK) {
     const int OC = dgrad ? C : K;
     const bool big = OC % 128 == 0;
     const bool nb3 = conv_nbuf() == 3;
@@ -415,28 +518,30 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
         if (big && nb3)
             launch_cfg<CONV_DGRAD, 128, 128, 2, 3>(src, wgt, out, N, H, W, C,
                                                    K, P, Q, R, S, sy, sx, py,
-                                                   px, stream);
+                                                   px, stream, nullptr,
+                                                   skpart, SK);
         else if (big)
             launch_cfg<CONV_DGRAD, 128, 128, 2>(src, wgt, out, N, H, W, C, K,
                                                 P, Q, R, S, sy, sx, py, px,
-                                                stream);
+                                                stream, nullptr, skpart, SK);
         else
             launch_cfg<CONV_DGRAD, 256, 64, 1>(src, wgt, out, N, H, W, C, K,
                                                P, Q, R, S, sy, sx, py, px,
-                                               stream);
+                                               stream, nullptr, skpart, SK);
     } else {
         if (big && nb3)
             launch_cfg<CONV_FWD, 128, 128, 2, 3>(src, wgt, out, N, H, W, C,
                                                  K, P, Q, R, S, sy, sx, py,
-                                                 px, stream, stats);
+                                                 px, stream, stats, skpart,
+                                                 SK);
         else if (big)
             launch_cfg<CONV_FWD, 128, 128, 2>(src, wgt, out, N, H, W, C, K,
                                               P, Q, R, S, sy, sx, py, px,
-                                              stream, stats);
+                                              stream, stats, skpart, SK);
         else
             launch_cfg<CONV_FWD, 256, 64, 1>(src, wgt, out, N, H, W, C, K,
                                              P, Q, R, S, sy, sx, py, px,
-                                             stream, stats);
+                                             stream, stats, skpart, SK);
     }
 }
 

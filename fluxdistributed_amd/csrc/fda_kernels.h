@@ -82,7 +82,14 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
                        bool dgrad, hipStream_t stream,
-                       float* stats = nullptr);
+                       float* stats = nullptr, float* skpart = nullptr,
+                       int SK = 1);
+
+// split-K combine: y = bf16(sum over SK fp32 partials) + optional BN
+// stats partials ([nblocks][2][OC])
+int conv_skcombine_blocks(long M, int OC);
+void conv_skcombine_launch(const float* part, void* y, float* stats, long M,
+                           int OC, int SK, int nblocks, hipStream_t stream);
 
 // stem conv (small C via channel-pad to 8, spatially pre-padded input)
 void conv_stem_fwd_launch(const void* src, const void* wgt, void* out,
