@@ -233,8 +233,11 @@ void adam_step(at::Tensor P, at::Tensor G, at::Tensor M, at::Tensor V,
 // ---- implicit-GEMM conv (NHWC bf16, conv_igemm.hip) -----------------------
 
 // split-K when the launch would underfill the 256-CU chip (layer4-size
-// shapes run at ~30% fill otherwise)
-static int conv_pick_sk(long M, int OC, int zbase) {
+// shapes ran at ~30% fill) AND the K-loop is deep enough to amortize the
+// fp32 partial round-trip (T < 48 measured a net loss: the 14x14 and
+// 1x1 shapes regressed up to 3x on the blanket trigger).
+static int conv_pick_sk(long M, int OC, int zbase, long T) {
+    if (T < 48) return 1;
     const int BM = (OC % 128 == 0) ? 128 : 256;
     const int BN = (OC % 128 == 0) ? 128 : 64;
     const long blocks = ((M + BM - 1) / BM) * (OC / BN) * zbase;
@@ -262,7 +265,7 @@ at::Tensor conv_igemm_fwd(at::Tensor x, at::Tensor w,
     auto y = at::empty({N, K, P, Q},
                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
     const long M = (long)N * P * Q;
-    const int SK = conv_pick_sk(M, K, /*zbase=*/1);
+    const int SK = conv_pick_sk(M, K, 1, (long)R * S * (C / 64));
     if (SK > 1) {
         auto part = at::empty({(long)SK * M * K},
                               x.options().dtype(at::kFloat));
@@ -300,7 +303,7 @@ std::tuple<at::Tensor, at::Tensor> conv_igemm_fwd_stats(
     const long M = (long)N * P * Q;
     auto y = at::empty({N, K, P, Q},
                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
-    const int SK = conv_pick_sk(M, K, 1);
+    const int SK = conv_pick_sk(M, K, 1, (long)R * S * (C / 64));
     if (SK > 1) {
         const int nb = fda::conv_skcombine_blocks(M, K);
         auto skp = at::empty({(long)SK * M * K}, x.options().dtype(at::kFloat));
@@ -343,7 +346,8 @@ at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
     const long M = (long)N * H * W;
     const int SK = (sy == 1 && sx == 1)
-        ? conv_pick_sk(M, (int)C, 1) : 1;   // parity classes already fan out
+        ? conv_pick_sk(M, (int)C, 1, (long)R * S * (K / 64))
+        : 1;                                // parity classes already fan out
     if (SK > 1) {
         auto skp = at::empty({(long)SK * M * C},
                              dy.options().dtype(at::kFloat));
