@@ -130,3 +130,18 @@ def check_nans(tree: GradTree) -> Dict[str, bool]:
         if v is not None and not torch.isfinite(v).all():
             bad[k] = True
     return bad
+
+
+def show_stats(tree: GradTree, name: str = "grads") -> str:
+    """Debug dump of per-leaf mean/std/min/max (the reference's
+    `_show_stats`, /root/reference/src/overloads.jl:56-59)."""
+    lines = [name]
+    for k, v in tree.items():
+        if v is None:
+            lines.append(f"  {k}: nothing")
+        else:
+            f = v.detach().float()
+            lines.append(
+                f"  {k}: shape={tuple(v.shape)} mean={f.mean():.4e} "
+                f"std={f.std():.4e} min={f.min():.4e} max={f.max():.4e}")
+    return "\n".join(lines)

@@ -111,3 +111,11 @@ def test_ensure_synced_and_nans():
     assert not ensure_synced([t1, t2])
     assert check_nans({"w": torch.tensor([1.0, float("nan")])}) == {"w": True}
     assert check_nans(t1) == {}
+
+
+def test_show_stats():
+    from fluxdistributed_amd.parallel.gradtree import show_stats
+
+    tree = {"a": torch.randn(4, 4), "b": None}
+    s = show_stats(tree)
+    assert "a: shape=(4, 4)" in s and "b: nothing" in s
