@@ -27,11 +27,25 @@
 // MFMA-per-glds ratio is what sets throughput, guide §5 ladder):
 //   OC % 128 == 0 : BM=128 x BN=128, 4 waves as 2x2, LDS 2x32 KiB
 //   OC % 128 != 0 : BM=256 x BN=64,  4 waves as 4x1, LDS 2x40 KiB
-// fp32 accumulate, bf16 store.
+// fp32 accumulate, bf16 store. Staging source offsets advance
+// incrementally within a filter tap (full address math only on tap
+// changes). The fwd epilogue can also emit per-channel sum/sumsq
+// partials of the rounded output for the downstream BatchNorm (`stats`),
+// and small-M deep-K shapes split the K loop over blockIdx.z into fp32
+// partials (`skpart`) folded by conv_skcombine_kernel — layer4-sized
+// grids otherwise fill only ~30% of the 256 CUs.
+//
+// This file also contains: the CONV_STEM mode (small-C stems on a
+// channel-padded C=8 image, r-only tap loop), the wgrad kernel
+// (reduction along the pixel axis via ds_read_tr16_b64 hardware
+// transpose reads from an m4-grouped LDS image, window positions
+// permuted for bank-conflict-free half-wave reads, fp32 atomic chunk
+// accumulation), and the batched weight transposer for the dgrad B
+// layout.
 //
 // Constraints (host wrapper): staged reduction channels (C fwd / K dgrad)
-// and output channels multiples of 64, dilation 1, groups 1. The ResNet
-// stem (C=3) falls back to the library path.
+// and output channels multiples of 64, dilation 1, groups 1; everything
+// else (grouped/dilated convs) falls back to the library path.
 
 #include <hip/hip_runtime.h>
 #include <cstdlib>
