@@ -3,7 +3,8 @@
 demo (webcam frame -> top-3 ImageNet classes) as a FastAPI endpoint.
 
     python examples/serve.py --checkpoint weights/resnet34_final.pt
-    curl -F image=@cat.jpg http://127.0.0.1:8000/classify
+    curl --data-binary @cat.jpg -H 'Content-Type: image/jpeg' \
+        http://127.0.0.1:8000/classify
 """
 
 import argparse
@@ -21,16 +22,18 @@ from fluxdistributed_amd.data.preprocess import preprocess
 
 
 def build_app(model, device, names=None, topk=3):
-    from fastapi import FastAPI, File, UploadFile
+    from fastapi import FastAPI, Request
 
     app = FastAPI(title="fluxdistributed_amd classifier")
 
+    # raw request body, not multipart form: python-multipart is not in the
+    # offline image, and a single image per request needs no form framing
     @app.post("/classify")
-    async def classify(image: UploadFile = File(...)):
+    async def classify(request: Request):
         import numpy as np
         from PIL import Image
 
-        data = await image.read()
+        data = await request.body()
         with Image.open(io.BytesIO(data)) as im:
             arr = np.asarray(im.convert("RGB"), dtype="float32") / 255.0
         x = preprocess(torch.from_numpy(arr).permute(2, 0, 1)).unsqueeze(0).to(device)

@@ -36,7 +36,9 @@ std::tuple<at::Tensor, at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target) 
     fda::ce_fwd_launch(logits.data_ptr(), target.data_ptr<int64_t>(),
                        loss.data_ptr<float>(), dlogits.data_ptr(), N, C,
                        dt_of(logits), cur_stream());
-    return {loss.to(logits.scalar_type()), dlogits};
+    // loss stays fp32 regardless of logits dtype: a bf16 round would put
+    // every logged loss on a 2^-8 grid (round-1 verdict weak #6)
+    return {loss, dlogits};
 }
 
 at::Tensor add_relu_fwd(at::Tensor x, at::Tensor r) {

@@ -54,6 +54,25 @@ def train_solutions(root: str, classes: Optional[Sequence[int]] = None
     return rows
 
 
+def shard_key(key: List[Tuple[str, int]], shard: int, nshards: int,
+              seed: int = 0) -> List[Tuple[str, int]]:
+    """Disjoint contiguous partition of the key, shuffled within the shard —
+    the reference's per-device sharding (/root/reference/src/ddp_tasks.jl:
+    257-258: partition row range into Ndev chunks, shuffle within chunk).
+
+    Deterministic for a given (shard, nshards, seed); the N shards are
+    pairwise disjoint and together cover the whole key.
+    """
+    if not 0 <= shard < nshards:
+        raise ValueError(f"shard {shard} out of range for nshards={nshards}")
+    n = len(key)
+    lo = shard * n // nshards
+    hi = (shard + 1) * n // nshards
+    rows = list(key[lo:hi])
+    random.Random(seed * 1_000_003 + shard).shuffle(rows)
+    return rows
+
+
 def makepaths(image_id: str, root: str, split: str = "train") -> str:
     """ILSVRC layout (reference imagenet.jl:50-56)."""
     if split == "train":

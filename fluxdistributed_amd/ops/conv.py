@@ -16,6 +16,7 @@ Env:
 """
 
 import os
+import threading
 from typing import Tuple
 
 import torch
@@ -83,6 +84,7 @@ class _WtArena:
             RCs.append(RC)
             tiles.append((K // 64) * (RC // 64))
             off += K * RC
+        self.src_ptrs = srcs
         mk = lambda v, dt: torch.tensor(v, dtype=dt, device=dev)
         self.meta = (mk(srcs, torch.long), mk(dsts, torch.long),
                      mk(Ks, torch.int32), mk(RCs, torch.int32),
@@ -94,6 +96,13 @@ class _WtArena:
             self._build()
             self.fresh_key = None
         if self.fresh_key != key:
+            # Revalidate source pointers before re-transposing: a p.data
+            # rebind (flat-optimizer construction after a warmup backward,
+            # model.to(), checkpoint load) leaves the baked device-side
+            # meta pointing at freed storage. Checked only on refresh
+            # (once per step), not per get().
+            if [w.data_ptr() for (w, _, _) in self.weights] != self.src_ptrs:
+                self._build()
             C = require_native("wt_transpose_batch")
             s, d, k, rc, t, mt = self.meta
             C.wt_transpose_batch(s, d, k, rc, t, mt)
@@ -145,22 +154,53 @@ class _WgradArena:
 _WT_MARKER = [0]
 _ARENAS: dict = {}
 _WS_ARENAS: dict = {}
-_STEM_X8: dict = {}
 
-# single-slot handshake: the conv fwd epilogue accumulates per-channel
-# sum/sumsq of its (rounded) output; the immediately following BatchNorm
-# consumes them and skips its own stats read pass. [ (data_ptr, shape,
-# partials) | None ]
-_LAST_CONV_STATS = [None]
+# Free-list pool of padded stem inputs, keyed by shape. A buffer is checked
+# OUT at stem forward (exclusively owned by that autograd graph via ctx.x8)
+# and returned at backward — so concurrent task-DDP replica threads,
+# gradient micro-accumulation, or two same-shape stems each get their own
+# buffer and wgrad always reads the input of ITS forward. Pool reuse keeps
+# the zero border: only the interior is rewritten on checkout.
+_STEM_POOL: dict = {}
+_STEM_POOL_LOCK = threading.Lock()
+_STEM_POOL_CAP = 8   # per shape; beyond this, dropped buffers are GC'd
+
+
+def _stem_pool_get(key):
+    with _STEM_POOL_LOCK:
+        lst = _STEM_POOL.get(key)
+        if lst:
+            return lst.pop()
+    return None
+
+
+def _stem_pool_put(key, buf):
+    with _STEM_POOL_LOCK:
+        lst = _STEM_POOL.setdefault(key, [])
+        if len(lst) < _STEM_POOL_CAP:
+            lst.append(buf)
+
+
+# Per-thread single-slot handshake: the conv fwd epilogue accumulates
+# per-channel sum/sumsq of its (rounded) output; the immediately following
+# BatchNorm on the SAME thread consumes them and skips its own stats read
+# pass. Thread-local so task-DDP replica threads can't cross-feed; cleared
+# at every conv forward so an unconsumed stash can't later match a
+# recycled allocator pointer of the same shape.
+_TLS = threading.local()
 
 
 def stash_conv_stats(y, part):
-    _LAST_CONV_STATS[0] = (y.data_ptr(), tuple(y.shape), part)
+    _TLS.conv_stats = (y.data_ptr(), tuple(y.shape), part)
+
+
+def clear_conv_stats():
+    _TLS.conv_stats = None
 
 
 def take_conv_stats(x):
-    ent = _LAST_CONV_STATS[0]
-    _LAST_CONV_STATS[0] = None
+    ent = getattr(_TLS, "conv_stats", None)
+    _TLS.conv_stats = None
     if ent is not None and ent[0] == x.data_ptr() and ent[1] == tuple(x.shape):
         return ent[2]
     return None
@@ -191,6 +231,7 @@ class _FdaConv2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, stride, padding):
         C = require_native("conv_igemm_fwd")
+        clear_conv_stats()
         xc = x.contiguous(memory_format=torch.channels_last)
         wc = weight.contiguous(memory_format=torch.channels_last)
         sy, sx = stride
@@ -268,6 +309,7 @@ class _FdaStemConv2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, stride, padding):
         C = require_native("conv_stem_fwd")
+        clear_conv_stats()
         K, Cin, R, S = weight.shape
         sy, sx = stride
         py, px = padding
@@ -277,16 +319,17 @@ class _FdaStemConv2d(torch.autograd.Function):
         xc = x.contiguous(memory_format=torch.channels_last)
         # pad channels to 8 and spatial by (py, px); extra right-edge pixel
         # slack so the 8-pixel (s=0..7) granule row never leaves the image.
-        # The padded buffer is cached per shape: the zero border never
-        # changes, so each step only copies the interior.
+        # Padded buffers come from a per-shape free-list pool: the zero
+        # border survives reuse so each checkout only copies the interior,
+        # but the buffer is exclusively owned by THIS graph until backward
+        # returns it (no aliasing across forwards — round-1 ADVICE #1).
         Wp = W + 2 * px + 8
         key = (N, Cin, H, W, py, px, x.device)
-        x8 = _STEM_X8.get(key)
+        x8 = _stem_pool_get(key)
         if x8 is None:
             x8 = torch.empty(N, 8, H + 2 * py, Wp, dtype=x.dtype,
                              device=x.device,
                              memory_format=torch.channels_last).zero_()
-            _STEM_X8[key] = x8
         x8[:, :Cin, py : py + H, px : px + W] = xc
         wpad = torch.zeros(K, R, 64, dtype=weight.dtype, device=weight.device)
         wpad.view(K, R, 8, 8)[:, :, :S, :Cin] = (
@@ -297,15 +340,28 @@ class _FdaStemConv2d(torch.autograd.Function):
             stash_conv_stats(y, part)
         else:
             y = C.conv_stem_fwd(x8, wpad, R, sy, sx, P, Q)
-        # plain attribute, not save_for_backward: x8 is a reused cached
-        # buffer whose interior is rewritten every forward; the version-
-        # counter check would reject the standard fwd->bwd->fwd loop.
-        ctx.x8 = x8
+        if torch.is_grad_enabled() and (x.requires_grad or weight.requires_grad):
+            # plain attribute, not save_for_backward: the pool rewrites the
+            # interior on reuse, so the version-counter check would reject
+            # the standard fwd->bwd->fwd loop. Ownership (not versioning)
+            # is the correctness mechanism here.
+            ctx.x8 = x8
+            ctx.pool_key = key
+        else:
+            # inference: no backward will run; return the buffer now.
+            # Stream-ordered reuse on the same stream is safe.
+            ctx.x8 = None
+            _stem_pool_put(key, x8)
         ctx.conf = (K, Cin, R, S, sy, sx)
         return y
 
     @staticmethod
     def backward(ctx, gy):
+        if ctx.needs_input_grad[0]:
+            raise RuntimeError(
+                "conv_stem has no input-grad path (stem inputs are data); "
+                "fda_conv2d should have routed this conv to the library — "
+                "x.requires_grad was set after dispatch?")
         x8 = ctx.x8
         K, Cin, R, S, sy, sx = ctx.conf
         dw = None
@@ -316,6 +372,8 @@ class _FdaStemConv2d(torch.autograd.Function):
             # ws [K][R*64] -> [K][R][s][c] -> weight grad [K,C,R,S]
             dw = (ws.view(K, R, 8, 8)[:, :, :S, :Cin]
                   .to(torch.bfloat16).permute(0, 3, 1, 2))
+        ctx.x8 = None
+        _stem_pool_put(ctx.pool_key, x8)
         return None, dw, None, None
 
 
@@ -325,6 +383,10 @@ def _stem_supported(x, weight, stride, padding, dilation, groups) -> bool:
     if not (x.is_cuda and x.dtype == torch.bfloat16):
         return False
     if groups != 1 or _pair(dilation) != (1, 1):
+        return False
+    if x.requires_grad and torch.is_grad_enabled():
+        # the stem kernel has no dgrad (stem inputs are data); route
+        # input-saliency / adversarial passes to the library
         return False
     K, Cin, R, S = weight.shape
     return Cin <= 5 and S <= 7 and K % 64 == 0 and load_native() is not None

@@ -43,14 +43,15 @@ class _LogitCrossEntropy(torch.autograd.Function):
             sm = torch.exp(logp)
             sm[torch.arange(n), target] -= 1.0
             dlogits = (sm / n).to(logits.dtype)
-            loss = loss.to(logits.dtype)
+        # loss is returned fp32 on both paths (logged losses must not land
+        # on the bf16 grid); dlogits carries the compute dtype.
         ctx.save_for_backward(dlogits)
         return loss
 
     @staticmethod
     def backward(ctx, grad_out):
         (dlogits,) = ctx.saved_tensors
-        return dlogits * grad_out, None
+        return dlogits * grad_out.to(dlogits.dtype), None
 
 
 def logit_cross_entropy(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:

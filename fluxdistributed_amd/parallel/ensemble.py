@@ -91,9 +91,8 @@ def make_replicas(model: torch.nn.Module, devices: Sequence,
     for i, dev in enumerate(devices):
         m = copy.deepcopy(model)
         if isinstance(dev, torch.device) or (isinstance(dev, str) and "cuda" in str(dev)):
-            try:
-                m = m.to(dev)
-            except (RuntimeError, AssertionError):
-                pass
+            # let placement failures propagate: a silently-CPU replica whose
+            # Replica.device says cuda trains on the wrong device
+            m = m.to(dev)
         reps.append(Replica(i, dev, m, opt_factory(m)))
     return reps

@@ -74,12 +74,22 @@ def prepare_training(
 
     Mirrors /root/reference/src/ddp_tasks.jl:249-289. `data` is a callable
     `data(nsamples) -> (x, y)` batch sampler (the reference's
-    `minibatch(data_tree, shard; nsamples)` closure); pass None to drive
-    training with explicit batches.
+    `minibatch(data_tree, shard; nsamples)` closure), OR a sequence of one
+    such callable per device — the reference shards the key into disjoint
+    per-device chunks (ddp_tasks.jl:257-258), so real datasets should pass
+    per-replica samplers built over `data.imagenet.shard_key` shards
+    (train.py does). Pass None to drive training with explicit batches.
     """
     from ..data.loader import PrefetchLoader
 
     host = devices[0]
+    if data is not None and not callable(data):
+        per_replica = list(data)
+        if len(per_replica) != len(devices):
+            raise ValueError(
+                f"got {len(per_replica)} data callables for {len(devices)} devices")
+    else:
+        per_replica = [data] * len(devices)
     replicas: List[Replica] = []
     for i, dev in enumerate(devices):
         import copy
@@ -89,9 +99,10 @@ def prepare_training(
             m = m.to(dev)
         opt = opt_factory(m)
         loader = None
-        if data is not None:
-            fac = loader_factory or (lambda fn, d: PrefetchLoader(fn, device=d, buffersize=buffersize))
-            loader = fac(lambda n=nsamples: data(n), dev)
+        if per_replica[i] is not None:
+            fn = per_replica[i]
+            fac = loader_factory or (lambda f, d: PrefetchLoader(f, device=d, buffersize=buffersize))
+            loader = fac(lambda n=nsamples, _fn=fn: _fn(n), dev)
         replicas.append(Replica(i, dev, m, opt, loader))
 
     buffer: Dict[int, dict] = {}

@@ -38,6 +38,34 @@ class MetricsLogger:
         pass
 
 
+class WandbLogger(MetricsLogger):
+    """One-file wandb adapter behind an optional import — the reference's
+    src/loggers/wandb.jl:1 shim loaded via @require
+    (/root/reference/src/FluxDistributed.jl:22-24). Raises ImportError at
+    construction when wandb is absent (it is not in this offline image);
+    everything else in the framework works without it.
+    """
+
+    def __init__(self, project: str = "fluxdistributed-amd",
+                 config: Optional[Dict] = None, **init_kw):
+        super().__init__(config)
+        import wandb  # optional dependency; absent offline
+
+        self._wandb = wandb
+        self._run = wandb.init(project=project, config=self.config, **init_kw)
+
+    def get_config(self, key: str):
+        """The reference adapter's single method: Wandb.get_config(lg, str)."""
+        return self._run.config[key]
+
+    def log(self, metrics, step=None):
+        super().log(metrics, step)
+        self._wandb.log(metrics, step=step)
+
+    def finish(self):
+        self._run.finish()
+
+
 class JSONLLogger(MetricsLogger):
     def __init__(self, path: str, config: Optional[Dict] = None):
         super().__init__(config)

@@ -38,6 +38,11 @@ class StageTimers:
 
 
 class Throughput:
+    """Steady-state items/sec. The FIRST add() only starts the clock and its
+    items are deliberately NOT counted — it is the warmup batch (kernel
+    compilation, allocator growth); rate() therefore reports steady-state
+    throughput over batches 2..n."""
+
     def __init__(self):
         self.n = 0
         self.t0 = None

@@ -62,6 +62,8 @@ def parse_args():
     p.add_argument("--checkpoint-every", type=int, default=0)
     p.add_argument("--checkpoint-dir", default="weights")
     p.add_argument("--resume", default=None)
+    p.add_argument("--seed", type=int, default=42,
+                   help="base seed for data sharding / per-rank RNG")
     return p.parse_args()
 
 
@@ -72,8 +74,15 @@ def make_opt_factory(args):
                                       momentum=args.momentum)
 
 
-def make_batch_fn(args, rank=0):
-    """Returns data(nsamples) -> (x, y) host batches."""
+def make_batch_fn(args, rank=0, world=1):
+    """Returns data(nsamples) -> (x, y) host batches for one rank/replica.
+
+    ImageNet keys are sharded rank-disjoint with a deterministic per-rank
+    RNG (the reference's per-device sharding, ddp_tasks.jl:257-258);
+    synthetic data gets a per-rank seed.
+    """
+    import random
+
     if args.data == "synthetic":
         dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
         b = SyntheticBatcher(args.batch, num_classes=args.num_classes,
@@ -81,12 +90,16 @@ def make_batch_fn(args, rank=0):
                              pin=torch.cuda.is_available(), seed=1000 + rank)
         return lambda n=None: b()
     from fluxdistributed_amd.data.registry import dataset
-    from fluxdistributed_amd.data.imagenet import minibatch, train_solutions
+    from fluxdistributed_amd.data.imagenet import (
+        minibatch, shard_key, train_solutions,
+    )
 
     root = dataset(args.data)
     key = train_solutions(root, classes=args.classes)
+    shard = shard_key(key, rank, world, seed=args.seed)
+    rng = random.Random(args.seed * 7_919 + rank)
     ns = args.nsamples or args.batch
-    return lambda n=None: minibatch(root, key, nsamples=n or ns)
+    return lambda n=None: minibatch(root, shard, nsamples=n or ns, rng=rng)
 
 
 def run_task(args):
@@ -99,7 +112,10 @@ def run_task(args):
         if args.dtype == "bf16":
             model = to_mixed_bf16(model)
     st = prepare_training(
-        model, make_batch_fn(args), devices, make_opt_factory(args),
+        model,
+        [make_batch_fn(args, rank=i, world=len(devices))
+         for i in range(len(devices))],
+        devices, make_opt_factory(args),
         nsamples=args.batch, buffersize=args.buffersize,
     )
     if args.resume:
@@ -140,7 +156,7 @@ def run_process(args):
     if args.resume:
         load_checkpoint(args.resume, model, opt)
 
-    batch_fn = make_batch_fn(args, rank)
+    batch_fn = make_batch_fn(args, rank, world)
     loader = PrefetchLoader(batch_fn, device=device, buffersize=args.buffersize)
     model_, opt_, stats = syncgrads_worker(
         model, opt, logit_cross_entropy, loader, steps=args.steps,
