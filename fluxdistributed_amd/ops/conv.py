@@ -60,15 +60,19 @@ class _WtArena:
         self.arena = None
         self.meta = None
         self.fresh_key = None
+        # arenas are shared per-device across task-DDP replica threads:
+        # register/rebuild/refresh are check-then-act and must not interleave
+        self.lock = threading.Lock()
 
     def register(self, weight):
-        if id(weight) in self.slices:
-            return
-        K, Cin, R, S = weight.shape
-        RC = R * S * Cin
-        self.weights.append((weight, K, RC))
-        self.slices[id(weight)] = (None, RC, K)
-        self.arena = None               # rebuild on next get
+        with self.lock:
+            if id(weight) in self.slices:
+                return
+            K, Cin, R, S = weight.shape
+            RC = R * S * Cin
+            self.weights.append((weight, K, RC))
+            self.slices[id(weight)] = (None, RC, K)
+            self.arena = None           # rebuild on next get
 
     def _build(self):
         dev = self.device
@@ -91,24 +95,25 @@ class _WtArena:
                      mk(tiles, torch.int32), max(tiles))
 
     def get(self, weight):
-        key = (_WT_MARKER[0], sum(w._version for (w, _, _) in self.weights))
-        if self.arena is None:
-            self._build()
-            self.fresh_key = None
-        if self.fresh_key != key:
-            # Revalidate source pointers before re-transposing: a p.data
-            # rebind (flat-optimizer construction after a warmup backward,
-            # model.to(), checkpoint load) leaves the baked device-side
-            # meta pointing at freed storage. Checked only on refresh
-            # (once per step), not per get().
-            if [w.data_ptr() for (w, _, _) in self.weights] != self.src_ptrs:
+        with self.lock:
+            key = (_WT_MARKER[0], sum(w._version for (w, _, _) in self.weights))
+            if self.arena is None:
                 self._build()
-            C = require_native("wt_transpose_batch")
-            s, d, k, rc, t, mt = self.meta
-            C.wt_transpose_batch(s, d, k, rc, t, mt)
-            self.fresh_key = key
-        off, RC, K = self.slices[id(weight)]
-        return self.arena[off : off + RC * K].view(RC, K)
+                self.fresh_key = None
+            if self.fresh_key != key:
+                # Revalidate source pointers before re-transposing: a p.data
+                # rebind (flat-optimizer construction after a warmup backward,
+                # model.to(), checkpoint load) leaves the baked device-side
+                # meta pointing at freed storage. Checked only on refresh
+                # (once per step), not per get().
+                if [w.data_ptr() for (w, _, _) in self.weights] != self.src_ptrs:
+                    self._build()
+                C = require_native("wt_transpose_batch")
+                s, d, k, rc, t, mt = self.meta
+                C.wt_transpose_batch(s, d, k, rc, t, mt)
+                self.fresh_key = key
+            off, RC, K = self.slices[id(weight)]
+            return self.arena[off : off + RC * K].view(RC, K)
 
 
 class _WgradArena:
@@ -125,30 +130,35 @@ class _WgradArena:
         self.arena = None
         self.zero_epoch = None
         self.seen = set()
+        # shared per-device across replica threads; without the lock two
+        # threads can BOTH see zero_epoch stale and the second whole-arena
+        # zero_() lands after the first thread's wgrad kernel (silent wipe)
+        self.lock = threading.Lock()
 
     def get(self, weight):
-        K, Cin, R, S = weight.shape
-        n = K * Cin * R * S
-        if id(weight) not in self.slices:
-            self.slices[id(weight)] = (sum(self.sizes), n)
-            self.sizes.append(n)
-            self.arena = None
-        if self.arena is None:
-            total = sum(self.sizes)
-            self.arena = torch.empty(total, dtype=torch.float32,
-                                     device=self.device)
-            self.zero_epoch = None
-        epoch = _WT_MARKER[0]
-        off, n = self.slices[id(weight)]
-        sl = self.arena[off : off + n]
-        if self.zero_epoch != epoch:
-            self.arena.zero_()
-            self.zero_epoch = epoch
-            self.seen = set()
-        elif id(weight) in self.seen:
-            sl.zero_()
-        self.seen.add(id(weight))
-        return sl.view(K, R * S * Cin)
+        with self.lock:
+            K, Cin, R, S = weight.shape
+            n = K * Cin * R * S
+            if id(weight) not in self.slices:
+                self.slices[id(weight)] = (sum(self.sizes), n)
+                self.sizes.append(n)
+                self.arena = None
+            if self.arena is None:
+                total = sum(self.sizes)
+                self.arena = torch.empty(total, dtype=torch.float32,
+                                         device=self.device)
+                self.zero_epoch = None
+            epoch = _WT_MARKER[0]
+            off, n = self.slices[id(weight)]
+            sl = self.arena[off : off + n]
+            if self.zero_epoch != epoch:
+                self.arena.zero_()
+                self.zero_epoch = epoch
+                self.seen = set()
+            elif id(weight) in self.seen:
+                sl.zero_()
+            self.seen.add(id(weight))
+            return sl.view(K, R * S * Cin)
 
 
 _WT_MARKER = [0]
