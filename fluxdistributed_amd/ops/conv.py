@@ -350,7 +350,11 @@ class _FdaStemConv2d(torch.autograd.Function):
             stash_conv_stats(y, part)
         else:
             y = C.conv_stem_fwd(x8, wpad, R, sy, sx, P, Q)
-        if torch.is_grad_enabled() and (x.requires_grad or weight.requires_grad):
+        # NOTE: torch.is_grad_enabled() is ALWAYS False inside
+        # Function.forward (autograd disables grad around it), so the keep
+        # decision uses requires_grad only. Under no_grad the ctx (and the
+        # buffer) are simply GC'd — a pool miss, not a leak.
+        if x.requires_grad or weight.requires_grad:
             # plain attribute, not save_for_backward: the pool rewrites the
             # interior on reuse, so the version-counter check would reject
             # the standard fwd->bwd->fwd loop. Ownership (not versioning)
@@ -358,8 +362,9 @@ class _FdaStemConv2d(torch.autograd.Function):
             ctx.x8 = x8
             ctx.pool_key = key
         else:
-            # inference: no backward will run; return the buffer now.
-            # Stream-ordered reuse on the same stream is safe.
+            # pure inference (frozen weights): no backward can reference
+            # this buffer; return it now. Stream-ordered reuse on the same
+            # stream is safe.
             ctx.x8 = None
             _stem_pool_put(key, x8)
         ctx.conf = (K, Cin, R, S, sy, sx)
