@@ -233,19 +233,8 @@ void adam_step(at::Tensor P, at::Tensor G, at::Tensor M, at::Tensor V,
 }  // namespace
 
 // ---- implicit-GEMM conv (NHWC bf16, conv_igemm.hip) -----------------------
-
-// split-K when the launch would underfill the 256-CU chip (layer4-size
-// shapes ran at ~30% fill) AND the K-loop is deep enough to amortize the
-// fp32 partial round-trip (T < 48 measured a net loss: the 14x14 and
-// 1x1 shapes regressed up to 3x on the blanket trigger).
-static int conv_pick_sk(long M, int OC, int zbase, long T) {
-    if (T < 48) return 1;
-    const int BM = (OC % 128 == 0) ? 128 : 256;
-    const int BN = (OC % 128 == 0) ? 128 : 64;
-    const long blocks = ((M + BM - 1) / BM) * (OC / BN) * zbase;
-    if (blocks >= 384) return 1;
-    return blocks >= 192 ? 2 : 4;
-}
+// Tile geometry and split-K come from fda::conv_igemm_plan (the same
+// function the launcher obeys), so workspace shapes always match.
 
 at::Tensor conv_igemm_fwd(at::Tensor x, at::Tensor w,
                           int64_t sy, int64_t sx, int64_t py, int64_t px) {
@@ -267,7 +256,8 @@ at::Tensor conv_igemm_fwd(at::Tensor x, at::Tensor w,
     auto y = at::empty({N, K, P, Q},
                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
     const long M = (long)N * P * Q;
-    const int SK = conv_pick_sk(M, K, 1, (long)R * S * (C / 64));
+    int BM, BN, SK;
+    fda::conv_igemm_plan(M, K, (long)R * S * (C / 64), 1, &BM, &BN, &SK);
     if (SK > 1) {
         auto part = at::empty({(long)SK * M * K},
                               x.options().dtype(at::kFloat));
@@ -305,7 +295,8 @@ std::tuple<at::Tensor, at::Tensor> conv_igemm_fwd_stats(
     const long M = (long)N * P * Q;
     auto y = at::empty({N, K, P, Q},
                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
-    const int SK = conv_pick_sk(M, K, 1, (long)R * S * (C / 64));
+    int BM, BN, SK;
+    fda::conv_igemm_plan(M, K, (long)R * S * (C / 64), 1, &BM, &BN, &SK);
     if (SK > 1) {
         const int nb = fda::conv_skcombine_blocks(M, K);
         auto skp = at::empty({(long)SK * M * K}, x.options().dtype(at::kFloat));
@@ -319,7 +310,6 @@ std::tuple<at::Tensor, at::Tensor> conv_igemm_fwd_stats(
                                    cur_stream());
         return {y, part};
     }
-    const int BM = (K % 128 == 0) ? 128 : 256;
     const long mtiles = (M + BM - 1) / BM;
     auto part = at::empty({mtiles, 2, K}, x.options().dtype(at::kFloat));
     fda::conv_igemm_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
@@ -347,9 +337,10 @@ at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
     auto dx = at::empty({N, C, H, W},
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
     const long M = (long)N * H * W;
-    const int SK = (sy == 1 && sx == 1)
-        ? conv_pick_sk(M, (int)C, 1, (long)R * S * (K / 64))
-        : 1;                                // parity classes already fan out
+    int BM, BN, SK = 1;
+    if (sy == 1 && sx == 1)                 // parity classes already fan out
+        fda::conv_igemm_plan(M, (int)C, (long)R * S * (K / 64), 1,
+                             &BM, &BN, &SK);
     if (SK > 1) {
         auto skp = at::empty({(long)SK * M * C},
                              dy.options().dtype(at::kFloat));

@@ -71,8 +71,14 @@ enum ConvMode { CONV_FWD = 0, CONV_DGRAD = 1, CONV_STEM = 2 };
 
 constexpr int BK = 64;
 
+// NW = waves per block = (BM/64)*(BN/64): 4 for the 128x128 / 256x64
+// configs (256 threads, 2 blocks/CU), 8 for the 256x128 big tile (512
+// threads, 1 block/CU — fewer glds per wave per K-step: 6 vs 8, same 32
+// MFMAs, so a higher MFMA:staging ratio on the deep-K layers).
 template <int MODE, int BM, int BN, int WN, int NBUF = 2>
-__global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
+__global__ __launch_bounds__((BM / 64) * (BN / 64) * 64,
+                             512 / ((BM / 64) * (BN / 64) * 64))
+void conv_igemm_kernel(
     const unsigned short* __restrict__ src,
     const unsigned short* __restrict__ wgt,
     unsigned short* __restrict__ out,
@@ -88,11 +94,12 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
                                  Small-M late layers fill only ~30% of the
                                  chip otherwise. */
     int SK) {
+    constexpr int NW = (BM / 64) * (BN / 64);   // waves per block
     constexpr int A_ELEMS = BM * BK;
     constexpr int B_ELEMS = BN * BK;
     constexpr int BUF_ELEMS = A_ELEMS + B_ELEMS;
-    constexpr int AI = BM / 32;          // A glds per wave per tile
-    constexpr int BI = BN / 32;          // B glds per wave per tile
+    constexpr int AI = BM / (8 * NW);    // A glds per wave per tile
+    constexpr int BI = BN / (8 * NW);    // B glds per wave per tile
 
     const int OC = (MODE == CONV_DGRAD) ? C : K;
     const int RC = (MODE == CONV_FWD) ? C : ((MODE == CONV_STEM) ? 64 : K);
@@ -125,7 +132,7 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
     const int wn = wid % WN;             // wave col (64-channel granularity)
 
     extern __shared__ unsigned short lds[];   // [NBUF][BUF_ELEMS]
-    constexpr int GPW = BM / 32 + BN / 32;    // glds per wave per tile
+    constexpr int GPW = AI + BI;              // glds per wave per tile
 
     // ---- per-lane staging descriptors ------------------------------------
     int a_row[AI];
@@ -277,7 +284,9 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
         // reading before it arrived here. With NBUF==3 the counted wait
         // leaves tile it+1's DMA in flight across the barrier (T4).
         if (NBUF > 2 && it + 1 < itN) {
-            if (GPW == 8)
+            if constexpr (GPW == 6)
+                asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+            else if constexpr (GPW == 8)
                 asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
             else
                 asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
@@ -373,7 +382,7 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
             #pragma unroll
             for (int ni = 0; ni < 4; ++ni) {
                 sf[(wid * 4 + ni) * 16 + fcol] = ssum[ni];
-                sf[256 + (wid * 4 + ni) * 16 + fcol] = sq[ni];
+                sf[NW * 64 + (wid * 4 + ni) * 16 + fcol] = sq[ni];
             }
         }
         __builtin_amdgcn_s_barrier();
@@ -382,9 +391,9 @@ __global__ __launch_bounds__(256, 2) void conv_igemm_kernel(
             #pragma unroll
             for (int ni = 0; ni < 4; ++ni) {
                 float s = ssum[ni], z = sq[ni];
-                for (int w2 = wid + WN; w2 < 4; w2 += WN) {
+                for (int w2 = wid + WN; w2 < NW; w2 += WN) {
                     s += sf[(w2 * 4 + ni) * 16 + fcol];
-                    z += sf[256 + (w2 * 4 + ni) * 16 + fcol];
+                    z += sf[NW * 64 + (w2 * 4 + ni) * 16 + fcol];
                 }
                 prow[ni * 16 + fcol] = s;
                 prow[OC + ni * 16 + fcol] = z;
@@ -404,6 +413,7 @@ static void launch_cfg(const void* src, const void* wgt, void* out,
         ? (long)N * P * Q
         : (long)N * ((H + sy - 1) / sy) * ((W + sx - 1) / sx);
     const unsigned zbase = (MODE != CONV_DGRAD) ? 1u : (unsigned)(sy * sx);
+    constexpr unsigned NTHREADS = (BM / 64) * (BN / 64) * 64;
     dim3 grid((unsigned)((M + BM - 1) / BM), (unsigned)(OC / BN),
               zbase * (unsigned)SK);
     const size_t shmem = NBUF * (BM * BK + BN * BK) * sizeof(unsigned short);
@@ -416,8 +426,9 @@ static void launch_cfg(const void* src, const void* wgt, void* out,
         }();
         (void)raised;
     }
-    hipLaunchKernelGGL((conv_igemm_kernel<MODE, BM, BN, WN, NBUF>), grid, dim3(256),
-                       shmem, stream, (const unsigned short*)src,
+    hipLaunchKernelGGL((conv_igemm_kernel<MODE, BM, BN, WN, NBUF>), grid,
+                       dim3(NTHREADS), shmem, stream,
+                       (const unsigned short*)src,
                        (const unsigned short*)wgt, (unsigned short*)out,
                        N, H, W, C, K, P, Q, R, S, sy, sx, py, px, stats,
                        skpart, SK);
@@ -520,14 +531,73 @@ static int conv_nbuf() {
     return v;
 }
 
+static int conv_bigtile() {
+    // 256x128 512-thread tile (8 waves, 6 glds/wave/K-step at the same 32
+    // MFMAs). A/B knob FLUXDIST_CONV_BIGTILE: 0 = never, 1 = whenever
+    // OC%128==0 and the grid still fills at 1 block/CU.
+    static int v = [] {
+        const char* e = getenv("FLUXDIST_CONV_BIGTILE");
+        return e ? atoi(e) : 0;
+    }();
+    return v;
+}
+
+// One place that decides tile geometry + split-K for a conv launch; the
+// torch bindings call this too so the stats/skpart workspace shapes always
+// match what the kernel will write.
+//   M   = output rows (per z-class for dgrad), OC = output channels,
+//   T   = K-loop depth in BK=64 steps (nR*nS*RC/64), zbase = sy*sx classes
+//         for strided dgrad else 1.
+// split-K only when the launch would underfill the 256-CU chip AND the
+// K-loop is deep enough to amortize the fp32 partial round-trip (T < 48
+// measured a net loss: 14x14 / 1x1 shapes regressed up to 3x on a blanket
+// trigger).
+void conv_igemm_plan(long M, int OC, long T, int zbase,
+                     int* bm, int* bn, int* sk) {
+    const bool big = OC % 128 == 0;
+    int BM = big ? 128 : 256;
+    int BN = big ? 128 : 64;
+    int SK = 1;
+    if (T >= 48) {
+        const long blocks = ((M + BM - 1) / BM) * (OC / BN) * zbase;
+        if (blocks < 192) SK = 4;
+        else if (blocks < 384) SK = 2;
+    }
+    if (big && SK == 1 && conv_bigtile()) {
+        const long blocks256 = ((M + 255) / 256) * (OC / 128) * zbase;
+        if (blocks256 >= 256) { BM = 256; BN = 128; }
+    }
+    *bm = BM; *bn = BN; *sk = SK;
+}
+
 void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
                        bool dgrad, hipStream_t stream, float* stats,
                        float* skpart, int SK) {
     const int OC = dgrad ? C : K;
-    const bool big = OC % 128 == 0;
     const bool nb3 = conv_nbuf() == 3;
+    const long Mv = !dgrad
+        ? (long)N * P * Q
+        : (long)N * ((H + sy - 1) / sy) * ((W + sx - 1) / sx);
+    const int zbase = dgrad ? sy * sx : 1;
+    const long RC = dgrad ? K : C;
+    // T here mirrors the bindings' SK trigger; the plan's BM/BN choice is
+    // what this function must obey so workspace shapes match
+    int BM, BN, SKp;
+    conv_igemm_plan(Mv, OC, (long)R * S * (RC / 64), zbase, &BM, &BN, &SKp);
+    const bool big = BN == 128;
+    if (BM == 256 && BN == 128) {
+        if (dgrad)
+            launch_cfg<CONV_DGRAD, 256, 128, 2, 2>(
+                src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
+                stream, nullptr, skpart, SK);
+        else
+            launch_cfg<CONV_FWD, 256, 128, 2, 2>(
+                src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
+                stream, stats, skpart, SK);
+        return;
+    }
     if (dgrad) {
         if (big && nb3)
             launch_cfg<CONV_DGRAD, 128, 128, 2, 3>(src, wgt, out, N, H, W, C,

@@ -85,6 +85,13 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        float* stats = nullptr, float* skpart = nullptr,
                        int SK = 1);
 
+// Tile/split-K plan for a conv launch — the single source of truth the
+// bindings use to size the stats/skpart workspaces. M = output rows (per
+// z-class for dgrad), OC = output channels, T = K-loop depth in BK=64
+// steps, zbase = sy*sx for strided dgrad else 1.
+void conv_igemm_plan(long M, int OC, long T, int zbase,
+                     int* bm, int* bn, int* sk);
+
 // split-K combine: y = bf16(sum over SK fp32 partials) + optional BN
 // stats partials ([nblocks][2][OC])
 int conv_skcombine_blocks(long M, int OC);

@@ -14,6 +14,7 @@ fused optimizer's G), and buckets are contiguous slices of it — zero
 packing copies.
 """
 
+import os
 import threading
 from typing import List
 
@@ -53,6 +54,14 @@ class GradBucketer:
         self.pg = process_group
         self.average = average
         self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
+        # FLUXDIST_DDP_FORCE=1: launch the collectives even at world 1.
+        # RCCL 2.26 refuses two ranks of one communicator on one GPU
+        # ("Duplicate GPU detected", tools/rccl_probe.py), so this is how
+        # the async stream-ordered RCCL branch (in-flight works, bucket
+        # launch order, wait ordering) is exercised inside a 1-GPU lease:
+        # a 1-rank all_reduce runs real RCCL kernels and is the identity.
+        self.force = os.environ.get("FLUXDIST_DDP_FORCE", "0") == "1"
+        self._active = (self.world > 1) or (self.force and dist.is_initialized())
         self.buckets: List[dict] = []
         self._build(optimizer, bucket_cap_mb)
         self._works: List = []
@@ -136,7 +145,7 @@ class GradBucketer:
     def _launch_ready_locked(self):
         while self._next_launch < len(self.buckets) and self._ready[self._next_launch]:
             b = self.buckets[self._next_launch]
-            if self.world > 1:
+            if self._active:
                 seg = b["flat"][b["lo"]:b["hi"]]
                 if seg.is_cuda and dist.get_backend(self.pg) == "gloo":
                     # gloo's async CUDA collectives stage through host
@@ -177,7 +186,7 @@ class GradBucketer:
     def allreduce_now(self):
         """One-shot (non-overlapped) all-reduce of every flat grad buffer —
         used when hooks are unavailable (e.g. inside a hipGraph capture)."""
-        if self.world <= 1:
+        if not self._active:
             return
         flats = {id(b["flat"]): b["flat"] for b in self.buckets}
         if any(f.is_cuda for f in flats.values()) \

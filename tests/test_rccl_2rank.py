@@ -150,3 +150,67 @@ def test_rccl_two_rank_replica_identity_and_solo_parity(overlap):
     for a, b in zip(ddp_masters, solo):
         assert torch.allclose(a, b, rtol=2e-2, atol=2e-3), \
             f"solo-parity drift: max|d|={float((a - b).abs().max())}"
+
+
+def _world1_worker(rank, results, overlap):
+    """world=1 RCCL with FLUXDIST_DDP_FORCE=1: every bucket's all_reduce is
+    a real RCCL kernel (stream-ordered, async works, launch ordering) and
+    the identity — params must match a no-DDP solo run bit-for-bit."""
+    import torch.distributed as dist
+
+    os.environ.update(RANK="0", WORLD_SIZE="1", MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT="29761", LOCAL_RANK="0",
+                      FLUXDIST_DDP_FORCE="1")
+    torch.cuda.set_device(0)
+    from fluxdistributed_amd.ops import logit_cross_entropy
+    from fluxdistributed_amd.parallel.process_ddp import DDPModel
+
+    try:
+        dist.init_process_group("nccl", rank=0, world_size=1)
+        model, opt = _build(1000)
+        ddp = DDPModel(model, opt, bucket_cap_mb=1.0, overlap=overlap)
+        assert ddp.bucketer._active, "force flag must activate collectives"
+        x, y = _shard(0)
+        for _ in range(STEPS):
+            loss = logit_cross_entropy(ddp(x), y)
+            opt.zero_grad()
+            loss.backward()
+            ddp.finalize_backward()
+            opt.step()
+        torch.cuda.synchronize()
+        results["ddp"] = [g.master.detach().cpu() for g in opt.groups
+                          if g.master is not None]
+    except Exception as e:  # noqa: BLE001
+        results["err"] = repr(e)
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("overlap", [False, True],
+                         ids=["flat_allreduce", "bucket_overlap"])
+def test_rccl_world1_collectives_are_identity(overlap):
+    import torch.multiprocessing as mp
+
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_world1_worker, args=(results, overlap), nprocs=1, join=True)
+    results = dict(results)
+    assert "err" not in results, results.get("err")
+
+    # solo oracle with the SAME shard (not the concatenated one)
+    from fluxdistributed_amd.ops import logit_cross_entropy
+
+    model, opt = _build(1000)
+    x, y = _shard(0)
+    for _ in range(STEPS):
+        loss = logit_cross_entropy(model(x), y)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+    torch.cuda.synchronize()
+    solo = [g.master.detach().cpu() for g in opt.groups if g.master is not None]
+    assert len(solo) == len(results["ddp"])
+    for a, b in zip(results["ddp"], solo):
+        assert torch.equal(a, b), \
+            f"world-1 RCCL path diverged from solo: max|d|={float((a-b).abs().max())}"
