@@ -236,6 +236,17 @@ void adam_step(at::Tensor P, at::Tensor G, at::Tensor M, at::Tensor V,
 // Tile geometry and split-K come from fda::conv_igemm_plan (the same
 // function the launcher obeys), so workspace shapes always match.
 
+// in-launch split-K: per-(m,n,zclass)-tile ticket counters, zeroed on the
+// stream ahead of the launch (guide Guideline 16)
+static unsigned* sk_tickets(const at::Tensor& ref, long ntiles,
+                            at::Tensor& keepalive, hipStream_t stream) {
+    keepalive = at::empty({ntiles}, ref.options().dtype(at::kInt));
+    unsigned* p = (unsigned*)keepalive.data_ptr<int>();
+    hipMemsetAsync(p, 0, sizeof(unsigned) * ntiles, stream);
+    return p;
+}
+
+
 at::Tensor conv_igemm_fwd(at::Tensor x, at::Tensor w,
                           int64_t sy, int64_t sx, int64_t py, int64_t px) {
     TORCH_CHECK(x.dim() == 4 && w.dim() == 4);
@@ -261,6 +272,17 @@ at::Tensor conv_igemm_fwd(at::Tensor x, at::Tensor w,
     if (SK > 1) {
         auto part = at::empty({(long)SK * M * K},
                               x.options().dtype(at::kFloat));
+        if (fda::conv_use_inlsk()) {
+            at::Tensor cnt_t;
+            unsigned* cnt = sk_tickets(x, ((M + BM - 1) / BM) * (K / BN),
+                                       cnt_t, cur_stream());
+            fda::conv_igemm_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                                   N, H, W, C, K, P, Q, R, S, (int)sy,
+                                   (int)sx, (int)py, (int)px, false,
+                                   cur_stream(), nullptr,
+                                   part.data_ptr<float>(), SK, cnt);
+            return y;
+        }
         fda::conv_igemm_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
                                N, H, W, C, K, P, Q, R, S, (int)sy, (int)sx,
                                (int)py, (int)px, false, cur_stream(), nullptr,
@@ -298,8 +320,23 @@ std::tuple<at::Tensor, at::Tensor> conv_igemm_fwd_stats(
     int BM, BN, SK;
     fda::conv_igemm_plan(M, K, (long)R * S * (C / 64), 1, &BM, &BN, &SK);
     if (SK > 1) {
-        const int nb = fda::conv_skcombine_blocks(M, K);
         auto skp = at::empty({(long)SK * M * K}, x.options().dtype(at::kFloat));
+        if (fda::conv_use_inlsk()) {
+            // reducer blocks write the stats partials like a SK=1 launch:
+            // [mtiles][2][K] with mtiles = grid.x
+            const long mtiles = (M + BM - 1) / BM;
+            auto part = at::empty({mtiles, 2, K}, x.options().dtype(at::kFloat));
+            at::Tensor cnt_t;
+            unsigned* cnt = sk_tickets(x, mtiles * (K / BN), cnt_t,
+                                       cur_stream());
+            fda::conv_igemm_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                                   N, H, W, C, K, P, Q, R, S, (int)sy,
+                                   (int)sx, (int)py, (int)px, false,
+                                   cur_stream(), part.data_ptr<float>(),
+                                   skp.data_ptr<float>(), SK, cnt);
+            return {y, part};
+        }
+        const int nb = fda::conv_skcombine_blocks(M, K);
         auto part = at::empty({nb, 2, K}, x.options().dtype(at::kFloat));
         fda::conv_igemm_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
                                N, H, W, C, K, P, Q, R, S, (int)sy, (int)sx,
@@ -344,6 +381,17 @@ at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
     if (SK > 1) {
         auto skp = at::empty({(long)SK * M * C},
                              dy.options().dtype(at::kFloat));
+        if (fda::conv_use_inlsk()) {
+            at::Tensor cnt_t;
+            unsigned* cnt = sk_tickets(dy, ((M + BM - 1) / BM) * (C / BN),
+                                       cnt_t, cur_stream());
+            fda::conv_igemm_launch(dy.data_ptr(), wt.data_ptr(),
+                                   dx.data_ptr(), N, (int)H, (int)W, (int)C,
+                                   K, P, Q, (int)R, (int)S, (int)sy, (int)sx,
+                                   (int)py, (int)px, true, cur_stream(),
+                                   nullptr, skp.data_ptr<float>(), SK, cnt);
+            return dx;
+        }
         fda::conv_igemm_launch(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                                N, (int)H, (int)W, (int)C, K, P, Q, (int)R,
                                (int)S, (int)sy, (int)sx, (int)py, (int)px,

@@ -90,10 +90,15 @@ void conv_igemm_kernel(
                                  BN reduce+finalize directly (the separate
                                  bn_stats read pass is skipped) */
     float* __restrict__ skpart, /* split-K: fp32 partial output
-                                 [SK][M][OC]; combine kernel reduces+casts.
+                                 [SK][M][OC]; reduced either by the
+                                 separate combine kernel or in-launch by
+                                 the last-arriving slice (cnt != null).
                                  Small-M late layers fill only ~30% of the
                                  chip otherwise. */
-    int SK) {
+    int SK,
+    unsigned* __restrict__ cnt /* in-launch combine tickets, one per
+                                 (m,n,zclass) tile, memset to 0 before the
+                                 launch; null = separate combine kernel */) {
     constexpr int NW = (BM / 64) * (BN / 64);   // waves per block
     constexpr int A_ELEMS = BM * BK;
     constexpr int B_ELEMS = BN * BK;
@@ -365,6 +370,71 @@ void conv_igemm_kernel(
             }
         }
     }
+    if (skpart != nullptr && cnt != nullptr) {
+        // ---- in-launch split-K seam (guide §6 Guideline 16, counter form):
+        // publish this slice's fp32 slab with an agent-scope release and
+        // take a ticket; the last-arriving slice re-reads all SK slabs,
+        // writes the bf16 tile and accumulates the BN stats partials.
+        // Saves the separate combine kernel's launch boundary + the slab
+        // round trip on shallow-K small-M shapes, which is why split-K can
+        // trigger there at all (the round-trip combine measured a net loss
+        // below T=48).
+        const long tileid =
+            ((long)zrest * gridDim.y + blockIdx.y) * gridDim.x + blockIdx.x;
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __syncthreads();
+        unsigned* flag = (unsigned*)lds;      // reuse the ONE shared array
+        if (threadIdx.x == 0) {
+            __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+            // restated wait: ROCm 7.2 drops the post-wbl2 vmcnt when its
+            // scoreboard says this wave has nothing outstanding
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            const unsigned t = __hip_atomic_fetch_add(
+                &cnt[tileid], 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            flag[0] = (t == (unsigned)SK - 1) ? 1u : 0u;
+        }
+        __syncthreads();
+        const bool im_last = flag[0] != 0;
+        __syncthreads();          // flag consumed before any lds reuse
+        if (!im_last) return;
+        if (threadIdx.x == 0)
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        __syncthreads();
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const long m = m0 + wm * 64 + mi * 16 + frow0 + j;
+                if (m >= M) continue;
+                long obase;
+                if (MODE != CONV_DGRAD) {
+                    obase = m * OC;
+                } else {
+                    const int ww = (int)(m % OW);
+                    const int hh = (int)((m / OW) % OH);
+                    const int n = (int)(m / ((long)OW * OH));
+                    obase = (((long)n * H + a + (long)sy * hh) * W + b +
+                             (long)sx * ww) * C;
+                }
+                const float* prow = skpart + m * OC + n0 + wn * 64;
+                unsigned short* orow = out + obase + n0 + wn * 64;
+                #pragma unroll
+                for (int ni = 0; ni < 4; ++ni) {
+                    float v = 0.f;
+                    for (int z = 0; z < SK; ++z)
+                        v += prow[(long)z * M * OC + ni * 16 + fcol];
+                    const unsigned short us = f32_to_bf16bits(v);
+                    orow[ni * 16 + fcol] = us;
+                    if (MODE != CONV_DGRAD && stats != nullptr) {
+                        const float vr = bf16bits_to_f32(us);
+                        ssum[ni] += vr;
+                        sq[ni] += vr * vr;
+                    }
+                }
+            }
+        }
+    }
+
     if (MODE != CONV_DGRAD && stats != nullptr) {
         // fold the four 16-row lane groups, then the waves sharing this
         // channel column, then write this m-tile's [2][OC] partial slice
@@ -407,7 +477,8 @@ static void launch_cfg(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
                        hipStream_t stream, float* stats = nullptr,
-                       float* skpart = nullptr, int SK = 1) {
+                       float* skpart = nullptr, int SK = 1,
+                       unsigned* cnt = nullptr) {
     const int OC = (MODE == CONV_DGRAD) ? C : K;
     const long M = (MODE != CONV_DGRAD)
         ? (long)N * P * Q
@@ -431,7 +502,7 @@ static void launch_cfg(const void* src, const void* wgt, void* out,
                        (const unsigned short*)src,
                        (const unsigned short*)wgt, (unsigned short*)out,
                        N, H, W, C, K, P, Q, R, S, sy, sx, py, px, stats,
-                       skpart, SK);
+                       skpart, SK, cnt);
 }
 
 // ---- split-K combine: y = bf16(sum_sk part) (+ BN stats partials) --------
@@ -534,10 +605,23 @@ static int conv_nbuf() {
 static int conv_bigtile() {
     // 256x128 512-thread tile (8 waves, 6 glds/wave/K-step at the same 32
     // MFMAs). A/B knob FLUXDIST_CONV_BIGTILE: 0 = never, 1 = whenever
-    // OC%128==0 and the grid still fills at 1 block/CU.
+    // OC%128==0 and the grid still fills at 1 block/CU. Default 0: the r2
+    // microbench measured it 3-13% SLOWER on every eligible leg (block-
+    // level overlap lost at 1 block/CU — same mechanism as NBUF=3,
+    // profiles/ab_bigtile.md).
     static int v = [] {
         const char* e = getenv("FLUXDIST_CONV_BIGTILE");
         return e ? atoi(e) : 0;
+    }();
+    return v;
+}
+
+bool conv_use_inlsk() {
+    // in-launch split-K combine (last-arriver seam) vs the separate
+    // combine kernel. FLUXDIST_CONV_INLSK=0 restores the round-trip path.
+    static bool v = [] {
+        const char* e = getenv("FLUXDIST_CONV_INLSK");
+        return !(e && e[0] == '0');
     }();
     return v;
 }
@@ -558,7 +642,11 @@ void conv_igemm_plan(long M, int OC, long T, int zbase,
     int BM = big ? 128 : 256;
     int BN = big ? 128 : 64;
     int SK = 1;
-    if (T >= 48) {
+    // With the in-launch seam the fp32 round trip is gone, so split-K can
+    // trigger on much shallower K loops (T>=12 vs 48 for the two-kernel
+    // combine, where shallow shapes measured a net loss).
+    const long Tmin = conv_use_inlsk() ? 12 : 48;
+    if (T >= Tmin) {
         const long blocks = ((M + BM - 1) / BM) * (OC / BN) * zbase;
         if (blocks < 192) SK = 4;
         else if (blocks < 384) SK = 2;
@@ -574,7 +662,7 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
                        bool dgrad, hipStream_t stream, float* stats,
-                       float* skpart, int SK) {
+                       float* skpart, int SK, unsigned* cnt) {
     const int OC = dgrad ? C : K;
     const bool nb3 = conv_nbuf() == 3;
     const long Mv = !dgrad
@@ -591,11 +679,11 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
         if (dgrad)
             launch_cfg<CONV_DGRAD, 256, 128, 2, 2>(
                 src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
-                stream, nullptr, skpart, SK);
+                stream, nullptr, skpart, SK, cnt);
         else
             launch_cfg<CONV_FWD, 256, 128, 2, 2>(
                 src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
-                stream, stats, skpart, SK);
+                stream, stats, skpart, SK, cnt);
         return;
     }
     if (dgrad) {
@@ -603,29 +691,33 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
             launch_cfg<CONV_DGRAD, 128, 128, 2, 3>(src, wgt, out, N, H, W, C,
                                                    K, P, Q, R, S, sy, sx, py,
                                                    px, stream, nullptr,
-                                                   skpart, SK);
+                                                   skpart, SK, cnt);
         else if (big)
             launch_cfg<CONV_DGRAD, 128, 128, 2>(src, wgt, out, N, H, W, C, K,
                                                 P, Q, R, S, sy, sx, py, px,
-                                                stream, nullptr, skpart, SK);
+                                                stream, nullptr, skpart, SK,
+                                                cnt);
         else
             launch_cfg<CONV_DGRAD, 256, 64, 1>(src, wgt, out, N, H, W, C, K,
                                                P, Q, R, S, sy, sx, py, px,
-                                               stream, nullptr, skpart, SK);
+                                               stream, nullptr, skpart, SK,
+                                               cnt);
     } else {
         if (big && nb3)
             launch_cfg<CONV_FWD, 128, 128, 2, 3>(src, wgt, out, N, H, W, C,
                                                  K, P, Q, R, S, sy, sx, py,
                                                  px, stream, stats, skpart,
-                                                 SK);
+                                                 SK, cnt);
         else if (big)
             launch_cfg<CONV_FWD, 128, 128, 2>(src, wgt, out, N, H, W, C, K,
                                               P, Q, R, S, sy, sx, py, px,
-                                              stream, stats, skpart, SK);
+                                              stream, stats, skpart, SK,
+                                              cnt);
         else
             launch_cfg<CONV_FWD, 256, 64, 1>(src, wgt, out, N, H, W, C, K,
                                              P, Q, R, S, sy, sx, py, px,
-                                             stream, stats, skpart, SK);
+                                             stream, stats, skpart, SK,
+                                             cnt);
     }
 }
 

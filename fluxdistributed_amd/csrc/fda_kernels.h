@@ -83,7 +83,11 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int R, int S, int sy, int sx, int py, int px,
                        bool dgrad, hipStream_t stream,
                        float* stats = nullptr, float* skpart = nullptr,
-                       int SK = 1);
+                       int SK = 1, unsigned* cnt = nullptr);
+
+// true when split-K uses the in-launch last-arriver combine (cnt tickets)
+// instead of the separate conv_skcombine kernel (FLUXDIST_CONV_INLSK).
+bool conv_use_inlsk();
 
 // Tile/split-K plan for a conv launch — the single source of truth the
 // bindings use to size the stats/skpart workspaces. M = output rows (per
