@@ -504,6 +504,48 @@ at::Tensor conv_stem_wgrad(at::Tensor dy, at::Tensor x8, int64_t R,
     return ws;   // [K][R*64]; host slices [K][R][s<7][c<3]
 }
 
+at::Tensor pad_rows_bf16(at::Tensor src, int64_t ldl) {
+    // [M][C] bf16 -> new [M][ldl] with zero right-pad (FC-head padding)
+    TORCH_CHECK(src.dim() == 2 && src.is_contiguous() &&
+                src.scalar_type() == at::kBFloat16);
+    const long M = src.size(0);
+    const int C = (int)src.size(1);
+    TORCH_CHECK(C % 8 == 0 && ldl % 8 == 0 && ldl >= C);
+    auto dst = at::empty({M, ldl}, src.options());
+    fda::pad_rows_bf16_launch(dst.data_ptr(), src.data_ptr(), M, C, (int)ldl,
+                              cur_stream());
+    return dst;
+}
+
+void pad_rows_bf16_into(at::Tensor dst, at::Tensor src) {
+    TORCH_CHECK(dst.dim() == 2 && src.dim() == 2);
+    TORCH_CHECK(dst.is_contiguous() && src.is_contiguous());
+    TORCH_CHECK(dst.size(0) == src.size(0));
+    fda::pad_rows_bf16_launch(dst.data_ptr(), src.data_ptr(), src.size(0),
+                              (int)src.size(1), (int)dst.size(1),
+                              cur_stream());
+}
+
+void bias_add_rows_bf16(at::Tensor y, at::Tensor bias) {
+    TORCH_CHECK(y.dim() == 2 && y.is_contiguous() &&
+                y.scalar_type() == at::kBFloat16);
+    TORCH_CHECK(bias.numel() == y.size(1) && bias.is_contiguous() &&
+                bias.scalar_type() == at::kBFloat16);
+    fda::bias_add_rows_bf16_launch(y.data_ptr(), bias.data_ptr(), y.size(0),
+                                   (int)y.size(1), cur_stream());
+}
+
+void colsum_accum_bf16(at::Tensor g, at::Tensor dy) {
+    // g[c] += cast(sum_m dy[m][c]) for c < g.numel() (bias direct-grad)
+    TORCH_CHECK(dy.dim() == 2 && dy.is_contiguous() &&
+                dy.scalar_type() == at::kBFloat16);
+    TORCH_CHECK(g.is_contiguous() && g.scalar_type() == at::kBFloat16);
+    TORCH_CHECK(g.numel() <= dy.size(1));
+    fda::colsum_accum_bf16_launch(g.data_ptr(), dy.data_ptr(), dy.size(0),
+                                  (int)dy.size(1), (int)g.numel(),
+                                  cur_stream());
+}
+
 void grad_accum_bf16(at::Tensor g, at::Tensor ws) {
     TORCH_CHECK(g.scalar_type() == at::kBFloat16 && g.is_contiguous());
     TORCH_CHECK(ws.scalar_type() == at::kFloat && ws.is_contiguous());
@@ -575,6 +617,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "implicit-GEMM conv fwd (NHWC bf16, MFMA)");
     m.def("wt_transpose_batch", &wt_transpose_batch);
     m.def("grad_accum_bf16", &grad_accum_bf16);
+    m.def("pad_rows_bf16", &pad_rows_bf16);
+    m.def("pad_rows_bf16_into", &pad_rows_bf16_into);
+    m.def("bias_add_rows_bf16", &bias_add_rows_bf16);
+    m.def("colsum_accum_bf16", &colsum_accum_bf16);
     m.def("conv_stem_fwd", &conv_stem_fwd);
     m.def("conv_stem_wgrad", &conv_stem_wgrad);
     m.def("conv_igemm_wgrad_into", &conv_igemm_wgrad_into);

@@ -618,10 +618,15 @@ static int conv_bigtile() {
 
 bool conv_use_inlsk() {
     // in-launch split-K combine (last-arriver seam) vs the separate
-    // combine kernel. FLUXDIST_CONV_INLSK=0 restores the round-trip path.
+    // combine kernel. Default OFF: measured 2-2.6x SLOWER than the
+    // round-trip combine on every SK shape (r2 A/B, profiles/ab_inlsk.md)
+    // — the per-block agent-release fence (buffer_wbl2 L2 writeback) and
+    // the reducer's serial slab re-read dwarf the saved launch boundary at
+    // these 64 KB/tile slab sizes. Kept behind FLUXDIST_CONV_INLSK=1 for
+    // future shapes with small slabs.
     static bool v = [] {
         const char* e = getenv("FLUXDIST_CONV_INLSK");
-        return !(e && e[0] == '0');
+        return e && e[0] == '1';
     }();
     return v;
 }
@@ -642,11 +647,11 @@ void conv_igemm_plan(long M, int OC, long T, int zbase,
     int BM = big ? 128 : 256;
     int BN = big ? 128 : 64;
     int SK = 1;
-    // With the in-launch seam the fp32 round trip is gone, so split-K can
-    // trigger on much shallower K loops (T>=12 vs 48 for the two-kernel
-    // combine, where shallow shapes measured a net loss).
-    const long Tmin = conv_use_inlsk() ? 12 : 48;
-    if (T >= Tmin) {
+    // T >= 48 only: shallow-K shapes measured a net loss under split-K on
+    // BOTH combine flavors (round-trip r1.14; in-launch seam r2 A/B — the
+    // 4-6x slab traffic dominates at shallow T, these legs are
+    // traffic-bound, not fill-bound).
+    if (T >= 48) {
         const long blocks = ((M + BM - 1) / BM) * (OC / BN) * zbase;
         if (blocks < 192) SK = 4;
         else if (blocks < 384) SK = 2;

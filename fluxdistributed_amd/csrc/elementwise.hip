@@ -307,4 +307,86 @@ void grad_accum_bf16_launch(void* g, const float* ws, long n, hipStream_t s) {
                        (unsigned short*)g, ws, n);
 }
 
+// --------------------------------------------------------------------------
+// FC head support (ops/linear.py): the Dense layer runs on the conv
+// implicit-GEMM kernels with out_features padded to a multiple of 64
+// (conv_igemm requires K % 64 == 0; the reference's FC is 512->1000).
+// These three kernels keep that padding in-house — no aten copies.
+// --------------------------------------------------------------------------
+
+// dst[M][ldl] = src[M][C] with cols [C, ldl) zeroed (row padding)
+__global__ __launch_bounds__(256) void pad_rows_bf16_kernel(
+    unsigned short* __restrict__ dst, const unsigned short* __restrict__ src,
+    long M, int C, int ldl) {
+    const long nv = M * (long)(ldl / 8);
+    const int cv = C / 8, lv = ldl / 8;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+         i += (long)gridDim.x * blockDim.x) {
+        const long m = i / lv;
+        const int c8 = (int)(i % lv);
+        uint4 v = {0, 0, 0, 0};
+        if (c8 < cv) v = *(const uint4*)(src + m * C + c8 * 8);
+        *(uint4*)(dst + m * ldl + c8 * 8) = v;
+    }
+}
+
+void pad_rows_bf16_launch(void* dst, const void* src, long M, int C, int ldl,
+                          hipStream_t s) {
+    const long nv = M * (long)(ldl / 8);
+    dim3 grid((unsigned)((nv + 255) / 256 < 4096 ? (nv + 255) / 256 : 4096));
+    hipLaunchKernelGGL(pad_rows_bf16_kernel, grid, dim3(256), 0, s,
+                       (unsigned short*)dst, (const unsigned short*)src,
+                       M, C, ldl);
+}
+
+// y[M][ldl] += bias[ldl] (bias pre-padded with zeros)
+__global__ __launch_bounds__(256) void bias_add_rows_bf16_kernel(
+    unsigned short* __restrict__ y, const unsigned short* __restrict__ bias,
+    long M, int ldl) {
+    const long nv = M * (long)(ldl / 8);
+    const int lv = ldl / 8;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+         i += (long)gridDim.x * blockDim.x) {
+        const int c8 = (int)(i % lv);
+        unsigned short yv[8], bv[8];
+        *(uint4*)yv = *(const uint4*)(y + (i / lv) * ldl + c8 * 8);
+        *(uint4*)bv = *(const uint4*)(bias + c8 * 8);
+        #pragma unroll
+        for (int e = 0; e < 8; ++e)
+            yv[e] = f32_to_bf16bits(bf16bits_to_f32(yv[e]) +
+                                    bf16bits_to_f32(bv[e]));
+        *(uint4*)(y + (i / lv) * ldl + c8 * 8) = *(uint4*)yv;
+    }
+}
+
+void bias_add_rows_bf16_launch(void* y, const void* bias, long M, int ldl,
+                               hipStream_t s) {
+    const long nv = M * (long)(ldl / 8);
+    dim3 grid((unsigned)((nv + 255) / 256 < 4096 ? (nv + 255) / 256 : 4096));
+    hipLaunchKernelGGL(bias_add_rows_bf16_kernel, grid, dim3(256), 0, s,
+                       (unsigned short*)y, (const unsigned short*)bias, M,
+                       ldl);
+}
+
+// bias grad: g_bf16[c] += cast(sum_m dy[m][c]) for c < Cvalid.
+// One block covers 256 columns; row reads are lane-consecutive (coalesced).
+__global__ __launch_bounds__(256) void colsum_accum_bf16_kernel(
+    unsigned short* __restrict__ g, const unsigned short* __restrict__ dy,
+    long M, int ldl, int Cvalid) {
+    const int c = blockIdx.x * 256 + threadIdx.x;
+    if (c >= ldl) return;
+    float acc = 0.f;
+    for (long m = 0; m < M; ++m) acc += bf16bits_to_f32(dy[m * ldl + c]);
+    if (c < Cvalid)
+        g[c] = f32_to_bf16bits(bf16bits_to_f32(g[c]) + acc);
+}
+
+void colsum_accum_bf16_launch(void* g, const void* dy, long M, int ldl,
+                              int Cvalid, hipStream_t s) {
+    dim3 grid((unsigned)((ldl + 255) / 256));
+    hipLaunchKernelGGL(colsum_accum_bf16_kernel, grid, dim3(256), 0, s,
+                       (unsigned short*)g, (const unsigned short*)dy, M, ldl,
+                       Cvalid);
+}
+
 }  // namespace fda

@@ -126,6 +126,14 @@ void wt_transpose_launch(const long* src_ptrs, long* dst_ptrs, const int* Ks,
 // direct-grad flush: G_bf16 += cast(ws_f32), one conv-weight slice
 void grad_accum_bf16_launch(void* g, const float* ws, long n, hipStream_t s);
 
+// FC-head padding helpers (Dense on the conv kernels, ops/linear.py)
+void pad_rows_bf16_launch(void* dst, const void* src, long M, int C, int ldl,
+                          hipStream_t s);
+void bias_add_rows_bf16_launch(void* y, const void* bias, long M, int ldl,
+                               hipStream_t s);
+void colsum_accum_bf16_launch(void* g, const void* dy, long M, int ldl,
+                              int Cvalid, hipStream_t s);
+
 // flat fused optimizers. P: param dtype; M/V/S fp32; G param dtype.
 void sgd_step_launch(void* P, const void* G, float* M, float* V, int64_t n,
                      float lr, float mom, float wd, bool nesterov,
