@@ -14,6 +14,7 @@ from typing import List, Optional, Type
 import torch
 import torch.nn as nn
 
+from ..ops.linear import FdaLinear
 from ..ops.functional import (batch_norm_act, MaxPool2d,
                               GlobalAvgPool)
 
@@ -150,7 +151,7 @@ class ResNet(nn.Module):
         self.layer3 = self._make_layer(block, 256, layers[2], stride=2)
         self.layer4 = self._make_layer(block, 512, layers[3], stride=2)
         self.avgpool = GlobalAvgPool()
-        self.fc = nn.Linear(512 * block.expansion, num_classes)
+        self.fc = FdaLinear(512 * block.expansion, num_classes)
         self._init_weights()
 
     def _make_layer(self, block, cout, n, stride=1):

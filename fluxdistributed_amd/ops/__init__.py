@@ -21,3 +21,4 @@ from .functional import (  # noqa: F401
 )
 from .fused_optim import FusedSGDMomentum, FusedAdam  # noqa: F401
 from .conv import fda_conv2d, FdaConv2d  # noqa: F401
+from .linear import FdaLinear  # noqa: F401

@@ -50,13 +50,13 @@ class _FcPadArena:
             wpad = torch.zeros(Kp, Cin, dtype=weight.dtype,
                                device=weight.device)
             bpad = torch.zeros(Kp, dtype=weight.dtype, device=weight.device)
-            ent = dict(wpad=wpad, bpad=bpad, key=None, Kp=Kp)
+            # ONE canonical 4D channels_last view: the transpose arena keys
+            # by id(), so the same view object must be reused everywhere
+            wpad4 = wpad.view(Kp, 1, 1, Cin).permute(0, 3, 1, 2)
+            ent = dict(wpad=wpad, bpad=bpad, wpad4=wpad4, key=None, Kp=Kp)
             self.entries[id(weight)] = ent
-            # register the padded weight (as a 4D view) for the batched
-            # dgrad transpose: wt[(c)][k] with k contiguous
-            _arena_for(ent["wpad4"] if False else wpad.view(Kp, Cin, 1, 1)) \
-                .register(wpad.view(Kp, Cin, 1, 1))
-            ent["wpad4"] = wpad.view(Kp, Cin, 1, 1)
+            # register for the batched dgrad transpose: wt[c][k] k-contig
+            _arena_for(wpad4).register(wpad4)
         key = (_WT_MARKER[0], weight._version,
                bias._version if bias is not None else 0,
                weight.data_ptr())
@@ -64,7 +64,8 @@ class _FcPadArena:
             C = require_native("fc_pad")
             C.pad_rows_bf16_into(ent["wpad"], weight.detach().contiguous())
             if bias is not None:
-                ent["bpad"][: N] = bias.detach()
+                C.pad_rows_bf16_into(ent["bpad"].view(1, -1),
+                                     bias.detach().reshape(1, -1).contiguous())
             ent["key"] = key
         return ent
 
@@ -76,14 +77,12 @@ class _FdaLinear(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias):
         C = require_native("fc_fwd")
-        from .conv import _arena_for
 
         N, Cin = weight.shape
         ent = _FC_PAD.get(weight, bias)
         Kp = ent["Kp"]
         xc = x.contiguous()
-        y = C.conv_igemm_fwd(_as_nhwc_4d(xc),
-                             ent["wpad"].view(Kp, 1, 1, Cin).permute(0, 3, 1, 2),
+        y = C.conv_igemm_fwd(_as_nhwc_4d(xc), ent["wpad4"],
                              1, 1, 0, 0)          # [M, Kp, 1, 1]
         y2 = y.view(x.shape[0], Kp)
         if bias is not None:
@@ -139,7 +138,10 @@ def _fda_linear_supported(x, weight) -> bool:
     if not (x.is_cuda and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16 and x.dim() == 2):
         return False
-    return weight.shape[1] % 64 == 0 and load_native() is not None
+    # in_features staged as conv reduction channels (%64); out_features
+    # only needs %8 for the 16B-vector pad kernels (padded to %64 in-house)
+    return (weight.shape[1] % 64 == 0 and weight.shape[0] % 8 == 0
+            and load_native() is not None)
 
 
 class FdaLinear(torch.nn.Linear):
