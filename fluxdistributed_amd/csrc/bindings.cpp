@@ -359,7 +359,10 @@ std::tuple<at::Tensor, at::Tensor> conv_igemm_fwd_stats(
 at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
                             int64_t C, int64_t H, int64_t W,
                             int64_t R, int64_t S,
-                            int64_t sy, int64_t sx, int64_t py, int64_t px) {
+                            int64_t sy, int64_t sx, int64_t py, int64_t px,
+                            c10::optional<at::Tensor> accum) {
+    // accum: optional bf16 tensor in dx's layout added into the result in
+    // the epilogue (residual-junction grad fusion, ops/conv.py).
     // dy: [N,K,P,Q] channels_last; wt: [R*S*C, K] row-major (pre-transposed
     // weight, k contiguous). Output dx: [N,C,H,W] channels_last.
     TORCH_CHECK(dy.dim() == 4 && dy.scalar_type() == at::kBFloat16);
@@ -373,6 +376,14 @@ at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
                 "conv_igemm: C and K must be multiples of 64");
     auto dx = at::empty({N, C, H, W},
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+    const void* accp = nullptr;
+    if (accum.has_value() && accum->defined()) {
+        TORCH_CHECK(accum->is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                    accum->scalar_type() == at::kBFloat16 &&
+                    accum->numel() == dx.numel(),
+                    "dgrad accum tensor must be channels_last bf16 of dx's shape");
+        accp = accum->data_ptr();
+    }
     const long M = (long)N * H * W;
     int BM, BN, SK = 1;
     if (sy == 1 && sx == 1)                 // parity classes already fan out
@@ -381,7 +392,7 @@ at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
     if (SK > 1) {
         auto skp = at::empty({(long)SK * M * C},
                              dy.options().dtype(at::kFloat));
-        if (fda::conv_use_inlsk()) {
+        if (fda::conv_use_inlsk() && accp == nullptr) {
             at::Tensor cnt_t;
             unsigned* cnt = sk_tickets(dy, ((M + BM - 1) / BM) * (C / BN),
                                        cnt_t, cur_stream());
@@ -400,13 +411,13 @@ at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
         fda::conv_skcombine_launch(skp.data_ptr<float>(), dx.data_ptr(),
                                    nullptr, M, (int)C, SK,
                                    fda::conv_skcombine_blocks(M, (int)C),
-                                   cur_stream());
+                                   cur_stream(), accp);
         return dx;
     }
     fda::conv_igemm_launch(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                            N, (int)H, (int)W, (int)C, K, P, Q, (int)R, (int)S,
                            (int)sy, (int)sx, (int)py, (int)px, /*dgrad=*/true,
-                           cur_stream());
+                           cur_stream(), nullptr, nullptr, 1, nullptr, accp);
     return dx;
 }
 
@@ -627,6 +638,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_igemm_wgrad", &conv_igemm_wgrad,
           "implicit-GEMM conv weight-grad (NHWC bf16, MFMA + tr16 reads)");
     m.def("conv_igemm_dgrad", &conv_igemm_dgrad,
-          "implicit-GEMM conv input-grad (NHWC bf16, MFMA)");
+          "implicit-GEMM conv input-grad (NHWC bf16, MFMA; optional fused "
+          "+= accum epilogue)",
+          pybind11::arg("dy"), pybind11::arg("wt"), pybind11::arg("C"),
+          pybind11::arg("H"), pybind11::arg("W"), pybind11::arg("R"),
+          pybind11::arg("S"), pybind11::arg("sy"), pybind11::arg("sx"),
+          pybind11::arg("py"), pybind11::arg("px"),
+          pybind11::arg("accum") = pybind11::none());
     m.attr("_built_for") = "gfx950";
 }

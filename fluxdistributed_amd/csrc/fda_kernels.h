@@ -83,7 +83,8 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                        int R, int S, int sy, int sx, int py, int px,
                        bool dgrad, hipStream_t stream,
                        float* stats = nullptr, float* skpart = nullptr,
-                       int SK = 1, unsigned* cnt = nullptr);
+                       int SK = 1, unsigned* cnt = nullptr,
+                       const void* accsrc = nullptr);
 
 // true when split-K uses the in-launch last-arriver combine (cnt tickets)
 // instead of the separate conv_skcombine kernel (FLUXDIST_CONV_INLSK).
@@ -100,7 +101,8 @@ void conv_igemm_plan(long M, int OC, long T, int zbase,
 // stats partials ([nblocks][2][OC])
 int conv_skcombine_blocks(long M, int OC);
 void conv_skcombine_launch(const float* part, void* y, float* stats, long M,
-                           int OC, int SK, int nblocks, hipStream_t stream);
+                           int OC, int SK, int nblocks, hipStream_t stream,
+                           const void* accsrc = nullptr);
 
 // stem conv (small C via channel-pad to 8, spatially pre-padded input)
 void conv_stem_fwd_launch(const void* src, const void* wgt, void* out,

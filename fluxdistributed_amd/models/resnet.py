@@ -57,13 +57,31 @@ class FusedBNAct(nn.Module):
         self._nbt_pending = 0
         super()._load_from_state_dict(*args, **kw)
 
-    def forward(self, x: torch.Tensor, residual: Optional[torch.Tensor] = None):
+    def forward(self, x: torch.Tensor, residual: Optional[torch.Tensor] = None,
+                defer_gres: bool = False):
         if self.training:
             self._nbt_pending += 1
         return batch_norm_act(
             x, self.weight, self.bias, self.running_mean, self.running_var,
             self.training, self.momentum, self.eps, self.relu, residual,
+            defer_gres,
         )
+
+
+def _junction_fusible(x: torch.Tensor, conv: nn.Conv2d) -> bool:
+    """True when the identity-shortcut junction gradient can be folded into
+    `conv`'s native dgrad += epilogue: the SAME tensor x must reach the
+    native dgrad (channels_last, native-supported shape) and need a grad.
+    The contract is load-bearing — a deferred gres that no dgrad consumes
+    raises in ops/conv.py rather than silently dropping gradient."""
+    from ..ops.conv import _native_supported
+
+    if not (torch.is_grad_enabled() and x.requires_grad):
+        return False
+    if not (x.is_cuda and x.is_contiguous(memory_format=torch.channels_last)):
+        return False
+    return _native_supported(x, conv.weight, conv.stride, conv.padding,
+                             conv.dilation, conv.groups)
 
 
 def conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
@@ -91,10 +109,12 @@ class BasicBlock(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
-        identity = self.downsample(x) if self.downsample is not None else x
+        if self.downsample is not None:
+            out = self.bn1(self.conv1(x))
+            return self.bn2(self.conv2(out), residual=self.downsample(x))
         out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out), residual=identity)
-        return out
+        return self.bn2(self.conv2(out), residual=x,
+                        defer_gres=_junction_fusible(x, self.conv1))
 
 
 class Bottleneck(nn.Module):
@@ -112,11 +132,14 @@ class Bottleneck(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
-        identity = self.downsample(x) if self.downsample is not None else x
+        if self.downsample is not None:
+            out = self.bn1(self.conv1(x))
+            out = self.bn2(self.conv2(out))
+            return self.bn3(self.conv3(out), residual=self.downsample(x))
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out), residual=identity)
-        return out
+        return self.bn3(self.conv3(out), residual=x,
+                        defer_gres=_junction_fusible(x, self.conv1))
 
 
 class Downsample(nn.Module):

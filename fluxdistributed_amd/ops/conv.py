@@ -204,6 +204,42 @@ def stash_conv_stats(y, part):
     _TLS.conv_stats = (y.data_ptr(), tuple(y.shape), part)
 
 
+# Residual-junction grad stash: at an identity-shortcut junction the grad
+# wrt x is dgrad(conv1) + gres(BN residual). When the block knows conv1's
+# dgrad will run on the native kernel (models/resnet.py), the BN backward
+# DEFERS gres here (returning None to autograd) and the conv dgrad folds
+# it in via the kernel's += epilogue — autograd's separate
+# CUDAFunctor_add pass over the full activation disappears.
+# Thread-local (task-DDP replica threads), keyed by (data_ptr, shape) of
+# the junction tensor; entries MUST be consumed by the very next dgrad of
+# that tensor (backward visits bn2 ... conv1 strictly in order on the
+# identity path), and an unconsumed entry is a loud error, never a silent
+# gradient drop.
+
+
+def stash_junction_gres(key, gres):
+    d = getattr(_TLS, "junction", None)
+    if d is None:
+        d = _TLS.junction = {}
+    if key in d:
+        raise RuntimeError(
+            "residual-junction gres stash collision: previous deferred "
+            "gradient was never consumed by a conv dgrad (fusion contract "
+            "broken — check models/resnet.py _junction_fusible gating)")
+    d[key] = gres
+
+
+def take_junction_gres(x):
+    d = getattr(_TLS, "junction", None)
+    if not d:
+        return None
+    return d.pop((x.data_ptr(), tuple(x.shape)), None)
+
+
+def junction_stash_empty() -> bool:
+    return not getattr(_TLS, "junction", None)
+
+
 def clear_conv_stats():
     _TLS.conv_stats = None
 
@@ -273,8 +309,11 @@ class _FdaConv2d(torch.autograd.Function):
             else:
                 # wt[(r*S+s)*C + c][k]: k-contiguous rows for the B tile
                 wt = w.permute(2, 3, 1, 0).reshape(R * S * Cin, K).contiguous()
+            # deferred residual-junction grad for this input, if the block
+            # routed one here: fused += in the dgrad epilogue
+            acc = take_junction_gres(x)
             dx = C.conv_igemm_dgrad(gy, wt, Cin, x.shape[2], x.shape[3],
-                                    R, S, sy, sx, py, px)
+                                    R, S, sy, sx, py, px, acc)
         if ctx.needs_input_grad[1]:
             K, Cin, R, S = w.shape
             if os.environ.get("FLUXDIST_WGRAD", "") != "miopen":

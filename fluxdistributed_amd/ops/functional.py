@@ -152,6 +152,7 @@ class _BNAct(torch.autograd.Function):
         eps: float,
         relu: bool,
         residual: Optional[torch.Tensor],
+        defer_gres: bool,
     ):
         C = require_native("batch_norm_act")
         from .conv import take_conv_stats
@@ -169,6 +170,13 @@ class _BNAct(torch.autograd.Function):
         ctx.relu = relu
         ctx.has_residual = residual is not None
         ctx.training = training
+        # junction fusion: instead of returning gres to autograd (which
+        # would add it to conv1's dx in a separate full-tensor pass), stash
+        # it for that conv's dgrad += epilogue. Only set by the block when
+        # it verified the consuming conv runs the native dgrad.
+        ctx.defer_gres = bool(defer_gres) and residual is not None and relu
+        if ctx.defer_gres:
+            ctx.res_key = (residual.data_ptr(), tuple(residual.shape))
         return out
 
     @staticmethod
@@ -201,7 +209,12 @@ class _BNAct(torch.autograd.Function):
             # d(out)/d(residual) = relu-mask * grad_out: a byproduct of the
             # bwd-apply kernel's own mask computation (gres output).
             gres = gres_k if ctx.relu else grad_out
-        return gx, gw, gb, None, None, None, None, None, None, gres
+        if ctx.defer_gres:
+            from .conv import stash_junction_gres
+
+            stash_junction_gres(ctx.res_key, gres)
+            gres = None     # the conv dgrad epilogue folds it into dx
+        return gx, gw, gb, None, None, None, None, None, None, gres, None
 
 
 def batch_norm_act(
@@ -215,16 +228,19 @@ def batch_norm_act(
     eps: float = 1e-5,
     relu: bool = True,
     residual: Optional[torch.Tensor] = None,
+    defer_gres: bool = False,
 ) -> torch.Tensor:
     """BN (train or eval stats) + optional residual add + optional ReLU.
 
     GPU: one fused native kernel pair (NHWC). CPU: composed PyTorch ops —
-    identical math, used as the oracle.
+    identical math, used as the oracle. `defer_gres` routes the residual
+    gradient into the consuming conv's dgrad += epilogue (caller must
+    guarantee that conv runs the native path on the SAME tensor).
     """
     if _on_gpu(x):
         return _BNAct.apply(
             x, weight, bias, running_mean, running_var,
-            training, momentum, eps, relu, residual,
+            training, momentum, eps, relu, residual, defer_gres,
         )
     # CPU reference path: compute in fp32 (handles mixed bf16-x/fp32-params)
     out = F.batch_norm(
