@@ -30,6 +30,13 @@ SHAPES = [
     (2, 128, 14, 14, 512, 1, 1),   # bottleneck expand 1x1
     (2, 256, 27, 27, 64, 1, 2),    # odd spatial 1x1 s2
     (2, 64, 56, 56, 256, 1, 1),    # bottleneck downsample-free expand
+    # ResNet-50/152 bottleneck channel extremes (round-1 verdict next #7):
+    (2, 1024, 14, 14, 256, 1, 1),  # r50 layer3 reduce 1x1 C=1024
+    (2, 2048, 7, 7, 512, 1, 1),    # r50/152 layer4 reduce 1x1 C=2048
+    (96, 512, 7, 7, 2048, 1, 1),   # layer4 expand 1x1 K=2048 at full batch
+                                   # (M=4704 -> wgrad takes the FT=4 tile)
+    (2, 1024, 14, 14, 2048, 1, 2), # layer4 downsample 1x1 s2 K=2048
+    (4, 512, 14, 14, 512, 3, 2),   # r50 layer4 3x3 s2 C=K=512
 ]
 
 
