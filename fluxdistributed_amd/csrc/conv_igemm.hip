@@ -954,7 +954,15 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
         __builtin_amdgcn_s_barrier();
-        if (it + RING - 1 < nsteps) stage((it + RING - 1) % RING);
+        // NOTE: the next tile's stage() is issued AFTER the tr reads below.
+        // hipcc (ROCm 7.2) conservatively emits `s_waitcnt vmcnt(0)` before
+        // every ds_read_b64_tr_b16 cluster (the intrinsic is ordered
+        // against pending LDS-DMA writes it cannot alias-analyze), so a
+        // stage issued before the reads is fully drained before any MFMA
+        // runs — the r1 kernel had ZERO DMA/compute overlap here (the
+        // measured WAIT_INST 38.6%). Reads-then-stage makes the inserted
+        // drain wait only for data this iteration needs anyway, and the
+        // DMA now flies over the MFMA cluster.
         const unsigned short* buf = lds + (it % RING) * 2 * TILE_ELEMS;
         short4_ a[FT][2][2], b[FT][2][2];   // [fi][ks][half]
         #pragma unroll
@@ -978,6 +986,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
                 b[fi][ks][1] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
                     (__attribute__((address_space(3))) short4_*)(pb + p1 * 64));
             }
+        if (it + RING - 1 < nsteps) stage((it + RING - 1) % RING);
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int ks = 0; ks < 2; ++ks)

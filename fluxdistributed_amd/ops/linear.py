@@ -62,7 +62,10 @@ class _FcPadArena:
                weight.data_ptr())
         if ent["key"] != key:
             C = require_native("fc_pad")
-            C.pad_rows_bf16_into(ent["wpad"], weight.detach().contiguous())
+            # weight pads along ROWS: rows are contiguous, so the [N,Cin]
+            # -> [Kp,Cin] row pad is a flat [1, N*Cin] -> [1, Kp*Cin] pad
+            C.pad_rows_bf16_into(ent["wpad"].view(1, -1),
+                                 weight.detach().contiguous().view(1, -1))
             if bias is not None:
                 C.pad_rows_bf16_into(ent["bpad"].view(1, -1),
                                      bias.detach().reshape(1, -1).contiguous())
