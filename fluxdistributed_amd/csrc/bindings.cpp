@@ -28,14 +28,17 @@ std::pair<int64_t, int64_t> nhwc_rows(const at::Tensor& x) {
 }
 
 std::tuple<at::Tensor, at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target) {
-    TORCH_CHECK(logits.dim() == 2 && logits.is_contiguous());
+    // accepts row-strided logits (stride(1)==1): the FC head's %64-padded
+    // buffer is read in place, no contiguity copy
+    TORCH_CHECK(logits.dim() == 2 && logits.stride(1) == 1 &&
+                logits.stride(0) >= logits.size(1));
     TORCH_CHECK(target.scalar_type() == at::kLong && target.is_contiguous());
     const int N = (int)logits.size(0), C = (int)logits.size(1);
     auto loss = at::zeros({}, logits.options().dtype(at::kFloat));
-    auto dlogits = at::empty_like(logits);
+    auto dlogits = at::empty({N, C}, logits.options());
     fda::ce_fwd_launch(logits.data_ptr(), target.data_ptr<int64_t>(),
                        loss.data_ptr<float>(), dlogits.data_ptr(), N, C,
-                       dt_of(logits), cur_stream());
+                       (int)logits.stride(0), dt_of(logits), cur_stream());
     // loss stays fp32 regardless of logits dtype: a bf16 round would put
     // every logged loss on a 2^-8 grid (round-1 verdict weak #6)
     return {loss, dlogits};

@@ -305,19 +305,24 @@ void conv_igemm_kernel(
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
         __builtin_amdgcn_s_barrier();
-        if (it + NBUF - 1 < itN) stage((it + NBUF - 1) % NBUF, it + NBUF - 1);
+        // fragment reads are issued BEFORE the next tile's stage(): the
+        // ds_read latency then hides under stage()'s ~40 address-math and
+        // glds-issue instructions instead of stalling at the lgkmcnt(0)
+        // hipcc emits right before the first MFMA. (The buffers are
+        // disjoint: reads target it%NBUF, the stage targets
+        // (it+NBUF-1)%NBUF.)
         const unsigned short* buf = lds + (it % NBUF) * BUF_ELEMS;
         short8 af[4][2], bf[4][2];
         #pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
+        for (int kh = 0; kh < 2; ++kh) {
             #pragma unroll
-            for (int kh = 0; kh < 2; ++kh)
+            for (int mi = 0; mi < 4; ++mi)
                 af[mi][kh] = *(const short8*)(buf + a_off[mi][kh]);
-        #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
             #pragma unroll
-            for (int kh = 0; kh < 2; ++kh)
+            for (int ni = 0; ni < 4; ++ni)
                 bf[ni][kh] = *(const short8*)(buf + b_off[ni][kh]);
+        }
+        if (it + NBUF - 1 < itN) stage((it + NBUF - 1) % NBUF, it + NBUF - 1);
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int kh = 0; kh < 2; ++kh)

@@ -23,13 +23,16 @@ template <typename T>
 __global__ void ce_fwd_kernel(const T* __restrict__ logits,
                               const int64_t* __restrict__ target,
                               float* __restrict__ loss, T* __restrict__ dlogits,
-                              int N, int C) {
+                              int N, int C, int ldl) {
+    // ldl: logits row stride in elements (>= C). The FC head computes
+    // logits in a %64-padded buffer (ops/linear.py), so CE reads the
+    // padded rows in place instead of forcing a contiguity copy.
     extern __shared__ __attribute__((aligned(16))) float smem[];
     float* row = smem;            // C floats
     float* scratch = smem + C;    // block-reduce scratch (<= 16 floats)
 
     for (int n = blockIdx.x; n < N; n += gridDim.x) {
-        const T* xrow = logits + (int64_t)n * C;
+        const T* xrow = logits + (int64_t)n * ldl;
         T* drow = dlogits + (int64_t)n * C;
         const int t = (int)target[n];
 
@@ -61,17 +64,18 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
 }
 
 void ce_fwd_launch(const void* logits, const int64_t* target, float* loss,
-                   void* dlogits, int N, int C, DT dt, hipStream_t s) {
+                   void* dlogits, int N, int C, int ldl, DT dt,
+                   hipStream_t s) {
     dim3 grid(N < 2048 ? N : 2048), block(256);
     size_t shmem = (C + 16) * sizeof(float);
     if (dt == DT::BF16)
         hipLaunchKernelGGL(ce_fwd_kernel<unsigned short>, grid, block, shmem, s,
                            (const unsigned short*)logits, target, loss,
-                           (unsigned short*)dlogits, N, C);
+                           (unsigned short*)dlogits, N, C, ldl);
     else
         hipLaunchKernelGGL(ce_fwd_kernel<float>, grid, block, shmem, s,
                            (const float*)logits, target, loss, (float*)dlogits,
-                           N, C);
+                           N, C, ldl);
 }
 
 // --------------------------------------------------------------------------

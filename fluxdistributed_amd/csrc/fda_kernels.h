@@ -10,7 +10,8 @@ enum class DT { F32 = 0, BF16 = 1 };
 
 // fused logit cross-entropy: mean loss over N rows + dlogits in one pass
 void ce_fwd_launch(const void* logits, const int64_t* target, float* loss,
-                   void* dlogits, int N, int C, DT dt, hipStream_t s);
+                   void* dlogits, int N, int C, int ldl, DT dt,
+                   hipStream_t s);
 
 // out = relu(x + r)
 void add_relu_fwd_launch(const void* x, const void* r, void* out, int64_t n,
