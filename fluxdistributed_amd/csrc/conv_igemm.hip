@@ -305,12 +305,14 @@ void conv_igemm_kernel(
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
         __builtin_amdgcn_s_barrier();
-        // fragment reads are issued BEFORE the next tile's stage(): the
-        // ds_read latency then hides under stage()'s ~40 address-math and
-        // glds-issue instructions instead of stalling at the lgkmcnt(0)
-        // hipcc emits right before the first MFMA. (The buffers are
-        // disjoint: reads target it%NBUF, the stage targets
-        // (it+NBUF-1)%NBUF.)
+        // stage-before-reads: tried the other order in r2 (reads first so
+        // their latency hides under stage's address math) — measured
+        // +314 us/step on fwd+dgrad: delaying the glds issue shrinks the
+        // DMA's flight window over the MFMA phase, which costs more than
+        // the lgkm stall it saves. (The wgrad kernel is the opposite case:
+        // hipcc force-drains DMA before tr16 reads, so there reads-first
+        // is required — see conv_wgrad_kernel.)
+        if (it + NBUF - 1 < itN) stage((it + NBUF - 1) % NBUF, it + NBUF - 1);
         const unsigned short* buf = lds + (it % NBUF) * BUF_ELEMS;
         short8 af[4][2], bf[4][2];
         #pragma unroll
@@ -322,7 +324,6 @@ void conv_igemm_kernel(
             for (int ni = 0; ni < 4; ++ni)
                 bf[ni][kh] = *(const short8*)(buf + b_off[ni][kh]);
         }
-        if (it + NBUF - 1 < itN) stage((it + NBUF - 1) % NBUF, it + NBUF - 1);
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int kh = 0; kh < 2; ++kh)

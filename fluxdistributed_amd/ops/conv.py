@@ -125,7 +125,7 @@ class _WgradArena:
 
     def __init__(self, device):
         self.device = device
-        self.slices = {}               # id(weight) -> (offset, numel)
+        self.slices = {}               # id(weight) -> (offset, numel, wref)
         self.sizes = []
         self.arena = None
         self.zero_epoch = None
@@ -136,11 +136,17 @@ class _WgradArena:
         self.lock = threading.Lock()
 
     def get(self, weight):
+        import weakref
+
         with self.lock:
             K, Cin, R, S = weight.shape
             n = K * Cin * R * S
-            if id(weight) not in self.slices:
-                self.slices[id(weight)] = (sum(self.sizes), n)
+            ent = self.slices.get(id(weight))
+            if ent is not None and ent[2]() is not weight:
+                ent = None          # CPython recycled a dead weight's id
+            if ent is None:
+                self.slices[id(weight)] = (sum(self.sizes), n,
+                                           weakref.ref(weight))
                 self.sizes.append(n)
                 self.arena = None
             if self.arena is None:
@@ -149,7 +155,7 @@ class _WgradArena:
                                          device=self.device)
                 self.zero_epoch = None
             epoch = _WT_MARKER[0]
-            off, n = self.slices[id(weight)]
+            off, n, _ = self.slices[id(weight)]
             sl = self.arena[off : off + n]
             if self.zero_epoch != epoch:
                 self.arena.zero_()

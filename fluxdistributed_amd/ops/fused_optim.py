@@ -16,6 +16,7 @@ HBM-bound streaming, vectorized 16B/lane (guide Appendix B elementwise).
 """
 
 
+import weakref
 from typing import List
 
 import torch
@@ -31,9 +32,13 @@ def _bump_wt_marker():
     bump_conv_wt_marker()
 
 
-# id(param) -> (flat G tensor, offset, numel): lets backward ops write
-# gradients straight into the flat buffer (no AccumulateGrad kernel) —
-# see ops/conv.py and ops/functional.py "direct grad" paths.
+# id(param) -> (flat G tensor, offset, numel, weakref(param)): lets
+# backward ops write gradients straight into the flat buffer (no
+# AccumulateGrad kernel) — see ops/conv.py and ops/functional.py "direct
+# grad" paths. The weakref guards against CPython id reuse: entries are
+# never removed when an optimizer dies, and a NEW parameter allocated at
+# a recycled address must not alias the dead one's slice (silently wrong
+# gradients — caught by the r2 GPU suite).
 FLAT_SLICES = {}
 
 
@@ -42,7 +47,10 @@ def flat_grad_slice(param):
     ent = FLAT_SLICES.get(id(param))
     if ent is None:
         return None
-    G, off, n = ent
+    G, off, n, ref = ent
+    if ref() is not param:          # stale entry from a recycled id
+        del FLAT_SLICES[id(param)]
+        return None
     return G[off : off + n]
 
 
@@ -69,7 +77,7 @@ class _FlatGroup:
             v.copy_(p.data)
             p.data = v
             p.grad = self._view_like(self.G, off, p.data)
-            FLAT_SLICES[id(p)] = (self.G, off, p.numel())
+            FLAT_SLICES[id(p)] = (self.G, off, p.numel(), weakref.ref(p))
         self.master = self.P.float() if dt != torch.float32 else None
 
     @staticmethod

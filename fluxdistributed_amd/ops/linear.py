@@ -19,6 +19,8 @@ All padding/copy kernels are in-tree (csrc/elementwise.hip) — zero aten
 kernels in the FC path.
 """
 
+import weakref
+
 import torch
 import torch.nn.functional as F
 
@@ -44,6 +46,8 @@ class _FcPadArena:
         from .conv import _WT_MARKER, _arena_for
 
         ent = self.entries.get(id(weight))
+        if ent is not None and ent["wref"]() is not weight:
+            ent = None              # CPython recycled a dead weight's id
         N, Cin = weight.shape
         Kp = (N + 63) // 64 * 64
         if ent is None:
@@ -53,7 +57,8 @@ class _FcPadArena:
             # ONE canonical 4D channels_last view: the transpose arena keys
             # by id(), so the same view object must be reused everywhere
             wpad4 = wpad.view(Kp, 1, 1, Cin).permute(0, 3, 1, 2)
-            ent = dict(wpad=wpad, bpad=bpad, wpad4=wpad4, key=None, Kp=Kp)
+            ent = dict(wpad=wpad, bpad=bpad, wpad4=wpad4, key=None, Kp=Kp,
+                       wref=weakref.ref(weight))
             self.entries[id(weight)] = ent
             # register for the batched dgrad transpose: wt[c][k] k-contig
             _arena_for(wpad4).register(wpad4)

@@ -136,3 +136,69 @@ def test_cooperative_sentinel_shutdown():
     for p in procs:
         p.join(timeout=30)
     assert counts[0] == counts[1] == 5  # stopped after step index 4
+
+
+def _run_shard_worker(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from fluxdistributed_amd.parallel.process_ddp import DDPModel
+
+        model = _mlp(seed=100 + rank)   # divergent init: broadcast must fix
+        opt = FusedSGDMomentum(model.parameters(), lr=0.05, momentum=0.9)
+        ddp = DDPModel(model, opt, bucket_cap_mb=0.001, overlap=True)
+        g = torch.Generator().manual_seed(500 + rank)   # rank-disjoint shard
+        for _ in range(3):
+            x = torch.randn(4, 8, generator=g)
+            y = torch.randint(0, 4, (4,), generator=g)
+            loss = logit_cross_entropy(ddp(x), y)
+            opt.zero_grad()
+            loss.backward()
+            ddp.finalize_backward()
+            opt.step()
+        q.put((rank, [p.detach().numpy() for p in model.parameters()]))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"ERROR: {type(e).__name__}: {e}"))
+
+
+def test_world4_disjoint_shards_match_large_batch():
+    """world=4 with rank-disjoint shards == one solo run on the 4x batch
+    (grad averaging is math-equivalent to large-batch, ddp_tasks.jl:93-109)
+    — the replica-identity + solo-parity oracle beyond world 2."""
+    world = 4
+    port = 29800 + os.getpid() % 150
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_run_shard_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, payload = q.get(timeout=180)
+        assert not isinstance(payload, str), payload
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=30)
+
+    for r in range(1, world):
+        for a, b in zip(results[0], results[r]):
+            assert (a == b).all(), f"rank {r} diverged"
+
+    solo = _mlp(seed=100)               # rank0's broadcast init
+    opt = FusedSGDMomentum(solo.parameters(), lr=0.05, momentum=0.9)
+    gens = [torch.Generator().manual_seed(500 + r) for r in range(world)]
+    for _ in range(3):
+        xs, ys = [], []
+        for g in gens:
+            xs.append(torch.randn(4, 8, generator=g))
+            ys.append(torch.randint(0, 4, (4,), generator=g))
+        x, y = torch.cat(xs), torch.cat(ys)
+        opt.zero_grad()
+        logit_cross_entropy(solo(x), y).backward()
+        opt.step()
+    for p_solo, arr in zip(solo.parameters(), results[0]):
+        assert torch.allclose(p_solo.detach(), torch.from_numpy(arr),
+                              rtol=1e-4, atol=1e-5), "large-batch parity"
