@@ -560,6 +560,18 @@ void colsum_accum_bf16(at::Tensor g, at::Tensor dy) {
                                   cur_stream());
 }
 
+void grad_accum_batch(at::Tensor g_ptrs, at::Tensor ws_ptrs, at::Tensor ns,
+                      int64_t max_n) {
+    // device int64 tensors: g/ws addresses and lengths per slice
+    TORCH_CHECK(g_ptrs.is_cuda() && g_ptrs.scalar_type() == at::kLong);
+    TORCH_CHECK(ws_ptrs.numel() == g_ptrs.numel() &&
+                ns.numel() == g_ptrs.numel());
+    fda::grad_accum_batch_launch(g_ptrs.data_ptr<int64_t>(),
+                                 ws_ptrs.data_ptr<int64_t>(),
+                                 ns.data_ptr<int64_t>(),
+                                 (int)g_ptrs.numel(), max_n, cur_stream());
+}
+
 void grad_accum_bf16(at::Tensor g, at::Tensor ws) {
     TORCH_CHECK(g.scalar_type() == at::kBFloat16 && g.is_contiguous());
     TORCH_CHECK(ws.scalar_type() == at::kFloat && ws.is_contiguous());
@@ -631,6 +643,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "implicit-GEMM conv fwd (NHWC bf16, MFMA)");
     m.def("wt_transpose_batch", &wt_transpose_batch);
     m.def("grad_accum_bf16", &grad_accum_bf16);
+    m.def("grad_accum_batch", &grad_accum_batch);
     m.def("pad_rows_bf16", &pad_rows_bf16);
     m.def("pad_rows_bf16_into", &pad_rows_bf16_into);
     m.def("bias_add_rows_bf16", &bias_add_rows_bf16);

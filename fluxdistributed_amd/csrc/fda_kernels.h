@@ -128,6 +128,10 @@ void wt_transpose_launch(const long* src_ptrs, long* dst_ptrs, const int* Ks,
 
 // direct-grad flush: G_bf16 += cast(ws_f32), one conv-weight slice
 void grad_accum_bf16_launch(void* g, const float* ws, long n, hipStream_t s);
+// batched flush: all pending slices in one launch (device pointer arrays)
+void grad_accum_batch_launch(const long* g_ptrs, const long* ws_ptrs,
+                             const long* ns, int ntensors, long max_n,
+                             hipStream_t s);
 
 // FC-head padding helpers (Dense on the conv kernels, ops/linear.py)
 void pad_rows_bf16_launch(void* dst, const void* src, long M, int C, int ldl,

@@ -337,11 +337,12 @@ class _FdaConv2d(torch.autograd.Function):
                     ws = C.conv_igemm_wgrad(gy, x, R, S, sy, sx, py, px)
                 if g_sl is not None:
                     # direct grad: G slice memory order == ws order
-                    # ([K][R][S][C]); one fused cast+add, no AccumulateGrad
-                    C.grad_accum_bf16(g_sl, ws.reshape(-1))
-                    from ..parallel.bucketing import notify_grad_written
+                    # ([K][R][S][C]); one fused cast+add, no AccumulateGrad.
+                    # Queued + batched at backward end unless a DDP
+                    # bucketer needs it immediately (ops/gradflush.py).
+                    from .gradflush import queue_or_flush
 
-                    notify_grad_written(w)
+                    queue_or_flush(w, g_sl, ws.reshape(-1))
                     dw = None
                 else:
                     # ws [K][R*S*C] fp32 is exactly the channels_last weight

@@ -121,13 +121,22 @@ class _FdaLinear(torch.autograd.Function):
             dx4 = C.conv_igemm_dgrad(dyp4, wt, Cin, 1, 1, 1, 1, 1, 1, 0, 0)
             dx = dx4.permute(0, 2, 3, 1).reshape(M, Cin)
         if ctx.needs_input_grad[1]:
-            ws = C.conv_igemm_wgrad(dyp4, _as_nhwc_4d(x), 1, 1, 1, 1, 0, 0)
+            # step-scoped arena slice (pre-zeroed once per backward epoch)
+            # instead of a fresh at::zeros per step
+            from .conv import _WS_ARENAS, _WgradArena
+
+            dev = x.device
+            if dev not in _WS_ARENAS:
+                _WS_ARENAS[dev] = _WgradArena(dev)
+            ws = _WS_ARENAS[dev].get(ent["wpad4"])
+            C.conv_igemm_wgrad_into(dyp4, _as_nhwc_4d(x), ws, 1, 1, 1, 1, 0, 0)
             g_sl = flat_grad_slice(weight)
             if g_sl is not None:
-                C.grad_accum_bf16(g_sl, ws.reshape(-1)[: N * Cin])
-                notify_grad_written(weight)
+                from .gradflush import queue_or_flush
+
+                queue_or_flush(weight, g_sl, ws.reshape(-1)[: N * Cin])
             else:
-                dw = ws[:N].to(torch.bfloat16)
+                dw = ws.view(Kp, Cin)[:N].to(torch.bfloat16)
         if bias is not None and ctx.needs_input_grad[2]:
             gb_sl = flat_grad_slice(bias)
             if gb_sl is not None:
