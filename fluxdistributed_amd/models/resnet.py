@@ -74,8 +74,12 @@ def _junction_fusible(x: torch.Tensor, conv: nn.Conv2d) -> bool:
     native dgrad (channels_last, native-supported shape) and need a grad.
     The contract is load-bearing — a deferred gres that no dgrad consumes
     raises in ops/conv.py rather than silently dropping gradient."""
+    import os
+
     from ..ops.conv import _native_supported
 
+    if os.environ.get("FLUXDIST_JUNCTION", "1") == "0":
+        return False
     if not (torch.is_grad_enabled() and x.requires_grad):
         return False
     if not (x.is_cuda and x.is_contiguous(memory_format=torch.channels_last)):
