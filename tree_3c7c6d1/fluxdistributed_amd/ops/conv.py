@@ -1,0 +1,431 @@
+"""Conv2d dispatch: hand-written MFMA implicit-GEMM kernels with library
+fallback.
+
+The reference gets conv from cuDNN via NNlibCUDA; the
+MI355X-native path is fluxdistributed_amd/csrc/conv_igemm.hip —
+v_mfma_f32_16x16x32_bf16 tiles, LDS-staged via global_load_lds. Supported
+there: NHWC bf16, dilation 1, groups 1, C and K multiples of 64 (all
+ResNet body convs) — fwd, dgrad AND wgrad. Small-C stems (C<=5, e.g. the
+ImageNet 7x7 and CIFAR 3x3 stems) run on the dedicated CONV_STEM kernels;
+only genuinely unsupported shapes (grouped/dilated, odd channel counts)
+fall back to the library path.
+
+Env:
+  FLUXDIST_CONV=miopen   force the library path everywhere (A/B testing)
+  FLUXDIST_CONV=fda      force the native path (errors on unsupported)
+"""
+
+import os
+import threading
+from typing import Tuple
+
+import torch
+import torch.nn.functional as F
+
+from .native import load_native, require_native
+
+
+def _pair(v) -> Tuple[int, int]:
+    return (v, v) if isinstance(v, int) else tuple(v)
+
+
+def _native_supported(x: torch.Tensor, w: torch.Tensor, stride, padding,
+                      dilation, groups) -> bool:
+    if os.environ.get("FLUXDIST_CONV", "") == "miopen":
+        return False
+    if not (x.is_cuda and x.dtype == torch.bfloat16 and w.dtype == torch.bfloat16):
+        return False
+    if groups != 1 or _pair(dilation) != (1, 1):
+        return False
+    C, K = x.shape[1], w.shape[0]
+    if C % 64 != 0 or K % 64 != 0:
+        return False
+    return load_native() is not None
+
+
+class _WtArena:
+    """Per-device cache of transposed conv weights wt[rs*C+c][k] for dgrad.
+
+    All registered weights are transposed in ONE kernel launch per training
+    step (wt_transpose_batch) instead of 36 per-layer permutes. Freshness:
+    a marker bumped by the fused optimizers (ops/fused_optim.py write params
+    through raw kernels, invisible to autograd) plus the sum of the weights'
+    `_version` counters (covers torch.optim in-place updates).
+    """
+
+    def __init__(self, device):
+        self.device = device
+        self.weights = []               # [(weight, K, RC)]
+        self.slices = {}                # id(weight) -> (offset, RC, K)
+        self.arena = None
+        self.meta = None
+        self.fresh_key = None
+        # arenas are shared per-device across task-DDP replica threads:
+        # register/rebuild/refresh are check-then-act and must not interleave
+        self.lock = threading.Lock()
+
+    def register(self, weight):
+        with self.lock:
+            if id(weight) in self.slices:
+                return
+            K, Cin, R, S = weight.shape
+            RC = R * S * Cin
+            self.weights.append((weight, K, RC))
+            self.slices[id(weight)] = (None, RC, K)
+            self.arena = None           # rebuild on next get
+
+    def _build(self):
+        dev = self.device
+        total = sum(K * RC for (_, K, RC) in self.weights)
+        self.arena = torch.empty(total, dtype=torch.bfloat16, device=dev)
+        off = 0
+        srcs, dsts, Ks, RCs, tiles = [], [], [], [], []
+        for (w, K, RC) in self.weights:
+            self.slices[id(w)] = (off, RC, K)
+            srcs.append(w.data_ptr())
+            dsts.append(self.arena.data_ptr() + off * 2)
+            Ks.append(K)
+            RCs.append(RC)
+            tiles.append((K // 64) * (RC // 64))
+            off += K * RC
+        self.src_ptrs = srcs
+        mk = lambda v, dt: torch.tensor(v, dtype=dt, device=dev)
+        self.meta = (mk(srcs, torch.long), mk(dsts, torch.long),
+                     mk(Ks, torch.int32), mk(RCs, torch.int32),
+                     mk(tiles, torch.int32), max(tiles))
+
+    def get(self, weight):
+        with self.lock:
+            key = (_WT_MARKER[0], sum(w._version for (w, _, _) in self.weights))
+            if self.arena is None:
+                self._build()
+                self.fresh_key = None
+            if self.fresh_key != key:
+                # Revalidate source pointers before re-transposing: a p.data
+                # rebind (flat-optimizer construction after a warmup backward,
+                # model.to(), checkpoint load) leaves the baked device-side
+                # meta pointing at freed storage. Checked only on refresh
+                # (once per step), not per get().
+                if [w.data_ptr() for (w, _, _) in self.weights] != self.src_ptrs:
+                    self._build()
+                C = require_native("wt_transpose_batch")
+                s, d, k, rc, t, mt = self.meta
+                C.wt_transpose_batch(s, d, k, rc, t, mt)
+                self.fresh_key = key
+            off, RC, K = self.slices[id(weight)]
+            return self.arena[off : off + RC * K].view(RC, K)
+
+
+class _WgradArena:
+    """Step-scoped fp32 wgrad workspace: zeroed in ONE launch per backward
+    epoch instead of a per-layer `zeros` each (the wgrad kernel accumulates
+    into its slice with atomicAdd). A second wgrad call for the same weight
+    inside one epoch (gradient micro-accumulation) re-zeros just its slice
+    so the returned dw stays the per-backward gradient."""
+
+    def __init__(self, device):
+        self.device = device
+        self.slices = {}               # id(weight) -> (offset, numel)
+        self.sizes = []
+        self.arena = None
+        self.zero_epoch = None
+        self.seen = set()
+        # shared per-device across replica threads; without the lock two
+        # threads can BOTH see zero_epoch stale and the second whole-arena
+        # zero_() lands after the first thread's wgrad kernel (silent wipe)
+        self.lock = threading.Lock()
+
+    def get(self, weight):
+        with self.lock:
+            K, Cin, R, S = weight.shape
+            n = K * Cin * R * S
+            if id(weight) not in self.slices:
+                self.slices[id(weight)] = (sum(self.sizes), n)
+                self.sizes.append(n)
+                self.arena = None
+            if self.arena is None:
+                total = sum(self.sizes)
+                self.arena = torch.empty(total, dtype=torch.float32,
+                                         device=self.device)
+                self.zero_epoch = None
+            epoch = _WT_MARKER[0]
+            off, n = self.slices[id(weight)]
+            sl = self.arena[off : off + n]
+            if self.zero_epoch != epoch:
+                self.arena.zero_()
+                self.zero_epoch = epoch
+                self.seen = set()
+            elif id(weight) in self.seen:
+                sl.zero_()
+            self.seen.add(id(weight))
+            return sl.view(K, R * S * Cin)
+
+
+_WT_MARKER = [0]
+_ARENAS: dict = {}
+_WS_ARENAS: dict = {}
+
+# Free-list pool of padded stem inputs, keyed by shape. A buffer is checked
+# OUT at stem forward (exclusively owned by that autograd graph via ctx.x8)
+# and returned at backward — so concurrent task-DDP replica threads,
+# gradient micro-accumulation, or two same-shape stems each get their own
+# buffer and wgrad always reads the input of ITS forward. Pool reuse keeps
+# the zero border: only the interior is rewritten on checkout.
+_STEM_POOL: dict = {}
+_STEM_POOL_LOCK = threading.Lock()
+_STEM_POOL_CAP = 8   # per shape; beyond this, dropped buffers are GC'd
+
+
+def _stem_pool_get(key):
+    with _STEM_POOL_LOCK:
+        lst = _STEM_POOL.get(key)
+        if lst:
+            return lst.pop()
+    return None
+
+
+def _stem_pool_put(key, buf):
+    with _STEM_POOL_LOCK:
+        lst = _STEM_POOL.setdefault(key, [])
+        if len(lst) < _STEM_POOL_CAP:
+            lst.append(buf)
+
+
+# Per-thread single-slot handshake: the conv fwd epilogue accumulates
+# per-channel sum/sumsq of its (rounded) output; the immediately following
+# BatchNorm on the SAME thread consumes them and skips its own stats read
+# pass. Thread-local so task-DDP replica threads can't cross-feed; cleared
+# at every conv forward so an unconsumed stash can't later match a
+# recycled allocator pointer of the same shape.
+_TLS = threading.local()
+
+
+def stash_conv_stats(y, part):
+    _TLS.conv_stats = (y.data_ptr(), tuple(y.shape), part)
+
+
+def clear_conv_stats():
+    _TLS.conv_stats = None
+
+
+def take_conv_stats(x):
+    ent = getattr(_TLS, "conv_stats", None)
+    _TLS.conv_stats = None
+    if ent is not None and ent[0] == x.data_ptr() and ent[1] == tuple(x.shape):
+        return ent[2]
+    return None
+
+
+def _want_conv_stats() -> bool:
+    return (torch.is_grad_enabled()
+            and os.environ.get("FLUXDIST_BN_FUSE", "1") != "0")
+
+
+def bump_conv_wt_marker():
+    """Invalidate cached transposed weights. Called by the fused optimizers
+    after each raw-kernel parameter update."""
+    _WT_MARKER[0] += 1
+
+
+def _arena_for(weight) -> _WtArena:
+    dev = weight.device
+    if dev not in _ARENAS:
+        _ARENAS[dev] = _WtArena(dev)
+    return _ARENAS[dev]
+
+
+class _FdaConv2d(torch.autograd.Function):
+    """Forward + input-grad on the native implicit-GEMM kernels; weight-grad
+    via the library (aten convolution_backward with weight-only mask)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, padding):
+        C = require_native("conv_igemm_fwd")
+        clear_conv_stats()
+        xc = x.contiguous(memory_format=torch.channels_last)
+        wc = weight.contiguous(memory_format=torch.channels_last)
+        sy, sx = stride
+        py, px = padding
+        if _want_conv_stats():
+            y, part = C.conv_igemm_fwd_stats(xc, wc, sy, sx, py, px)
+            stash_conv_stats(y, part)
+        else:
+            y = C.conv_igemm_fwd(xc, wc, sy, sx, py, px)
+        ctx.save_for_backward(xc, wc)
+        ctx.conf = (stride, padding, weight is wc)
+        if weight is wc:  # already channels_last: cacheable by identity
+            _arena_for(wc).register(wc)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, w = ctx.saved_tensors
+        stride, padding, cacheable = ctx.conf
+        sy, sx = stride
+        py, px = padding
+        gy = gy.contiguous(memory_format=torch.channels_last)
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            C = require_native("conv_igemm_dgrad")
+            K, Cin, R, S = w.shape
+            if cacheable:
+                wt = _arena_for(w).get(w)
+            else:
+                # wt[(r*S+s)*C + c][k]: k-contiguous rows for the B tile
+                wt = w.permute(2, 3, 1, 0).reshape(R * S * Cin, K).contiguous()
+            dx = C.conv_igemm_dgrad(gy, wt, Cin, x.shape[2], x.shape[3],
+                                    R, S, sy, sx, py, px)
+        if ctx.needs_input_grad[1]:
+            K, Cin, R, S = w.shape
+            if os.environ.get("FLUXDIST_WGRAD", "") != "miopen":
+                from .fused_optim import flat_grad_slice
+
+                C = require_native("conv_igemm_wgrad")
+                g_sl = flat_grad_slice(w) if cacheable else None
+                if cacheable:
+                    dev = w.device
+                    if dev not in _WS_ARENAS:
+                        _WS_ARENAS[dev] = _WgradArena(dev)
+                    ws = _WS_ARENAS[dev].get(w)
+                    C.conv_igemm_wgrad_into(gy, x, ws, R, S, sy, sx, py, px)
+                else:
+                    ws = C.conv_igemm_wgrad(gy, x, R, S, sy, sx, py, px)
+                if g_sl is not None:
+                    # direct grad: G slice memory order == ws order
+                    # ([K][R][S][C]); one fused cast+add, no AccumulateGrad
+                    C.grad_accum_bf16(g_sl, ws.reshape(-1))
+                    from ..parallel.bucketing import notify_grad_written
+
+                    notify_grad_written(w)
+                    dw = None
+                else:
+                    # ws [K][R*S*C] fp32 is exactly the channels_last weight
+                    # memory order [K][R][S][C]: one flat cast, no-copy view
+                    dw = (ws.to(torch.bfloat16).view(K, R, S, Cin)
+                          .permute(0, 3, 1, 2))
+            else:
+                dw = torch.ops.aten.convolution_backward(
+                    gy, x, w, None, list(stride), list(padding), [1, 1],
+                    False, [0, 0], 1, [False, True, False])[1]
+        return dx, dw, None, None
+
+
+class _FdaStemConv2d(torch.autograd.Function):
+    """Small-C stem conv (C<=5, S<=7): channel-pad to 8 + spatial pre-pad,
+    then the CONV_STEM kernel (one K-step per filter row r — 8 pixels x 8
+    channels as 64 virtual reduction channels). No dgrad (the stem input is
+    data); wgrad via conv_stem_wgrad."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, padding):
+        C = require_native("conv_stem_fwd")
+        clear_conv_stats()
+        K, Cin, R, S = weight.shape
+        sy, sx = stride
+        py, px = padding
+        N, _, H, W = x.shape
+        P = (H + 2 * py - R) // sy + 1
+        Q = (W + 2 * px - S) // sx + 1
+        xc = x.contiguous(memory_format=torch.channels_last)
+        # pad channels to 8 and spatial by (py, px); extra right-edge pixel
+        # slack so the 8-pixel (s=0..7) granule row never leaves the image.
+        # Padded buffers come from a per-shape free-list pool: the zero
+        # border survives reuse so each checkout only copies the interior,
+        # but the buffer is exclusively owned by THIS graph until backward
+        # returns it (no aliasing across forwards — round-1 ADVICE #1).
+        Wp = W + 2 * px + 8
+        key = (N, Cin, H, W, py, px, x.device)
+        x8 = _stem_pool_get(key)
+        if x8 is None:
+            x8 = torch.empty(N, 8, H + 2 * py, Wp, dtype=x.dtype,
+                             device=x.device,
+                             memory_format=torch.channels_last).zero_()
+        x8[:, :Cin, py : py + H, px : px + W] = xc
+        wpad = torch.zeros(K, R, 64, dtype=weight.dtype, device=weight.device)
+        wpad.view(K, R, 8, 8)[:, :, :S, :Cin] = (
+            weight.contiguous(memory_format=torch.channels_last)
+            .permute(0, 2, 3, 1))  # [K][R][S][C]
+        if _want_conv_stats():
+            y, part = C.conv_stem_fwd_stats(x8, wpad, R, sy, sx, P, Q)
+            stash_conv_stats(y, part)
+        else:
+            y = C.conv_stem_fwd(x8, wpad, R, sy, sx, P, Q)
+        # NOTE: torch.is_grad_enabled() is ALWAYS False inside
+        # Function.forward (autograd disables grad around it), so the keep
+        # decision uses requires_grad only. Under no_grad the ctx (and the
+        # buffer) are simply GC'd — a pool miss, not a leak.
+        if x.requires_grad or weight.requires_grad:
+            # plain attribute, not save_for_backward: the pool rewrites the
+            # interior on reuse, so the version-counter check would reject
+            # the standard fwd->bwd->fwd loop. Ownership (not versioning)
+            # is the correctness mechanism here.
+            ctx.x8 = x8
+            ctx.pool_key = key
+        else:
+            # pure inference (frozen weights): no backward can reference
+            # this buffer; return it now. Stream-ordered reuse on the same
+            # stream is safe.
+            ctx.x8 = None
+            _stem_pool_put(key, x8)
+        ctx.conf = (K, Cin, R, S, sy, sx)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        if ctx.needs_input_grad[0]:
+            raise RuntimeError(
+                "conv_stem has no input-grad path (stem inputs are data); "
+                "fda_conv2d should have routed this conv to the library — "
+                "x.requires_grad was set after dispatch?")
+        x8 = ctx.x8
+        K, Cin, R, S, sy, sx = ctx.conf
+        dw = None
+        if ctx.needs_input_grad[1]:
+            C = require_native("conv_stem_wgrad")
+            gyc = gy.contiguous(memory_format=torch.channels_last)
+            ws = C.conv_stem_wgrad(gyc, x8, R, sy, sx)
+            # ws [K][R*64] -> [K][R][s][c] -> weight grad [K,C,R,S]
+            dw = (ws.view(K, R, 8, 8)[:, :, :S, :Cin]
+                  .to(torch.bfloat16).permute(0, 3, 1, 2))
+        ctx.x8 = None
+        _stem_pool_put(ctx.pool_key, x8)
+        return None, dw, None, None
+
+
+def _stem_supported(x, weight, stride, padding, dilation, groups) -> bool:
+    if os.environ.get("FLUXDIST_CONV", "") == "miopen":
+        return False
+    if not (x.is_cuda and x.dtype == torch.bfloat16):
+        return False
+    if groups != 1 or _pair(dilation) != (1, 1):
+        return False
+    if x.requires_grad and torch.is_grad_enabled():
+        # the stem kernel has no dgrad (stem inputs are data); route
+        # input-saliency / adversarial passes to the library
+        return False
+    K, Cin, R, S = weight.shape
+    return Cin <= 5 and S <= 7 and K % 64 == 0 and load_native() is not None
+
+
+def fda_conv2d(x: torch.Tensor, weight: torch.Tensor, stride=(1, 1),
+               padding=(0, 0), dilation=(1, 1), groups: int = 1) -> torch.Tensor:
+    """conv2d with per-shape dispatch to the native MFMA kernel."""
+    stride, padding, dilation = _pair(stride), _pair(padding), _pair(dilation)
+    if _native_supported(x, weight, stride, padding, dilation, groups):
+        return _FdaConv2d.apply(x, weight, stride, padding)
+    if _stem_supported(x, weight, stride, padding, dilation, groups):
+        return _FdaStemConv2d.apply(x, weight, stride, padding)
+    if os.environ.get("FLUXDIST_CONV", "") == "fda" and x.is_cuda:
+        raise RuntimeError(
+            f"FLUXDIST_CONV=fda but shape unsupported by conv_igemm: "
+            f"x={tuple(x.shape)} w={tuple(weight.shape)} groups={groups}")
+    return F.conv2d(x, weight, None, stride, padding, dilation, groups)
+
+
+class FdaConv2d(torch.nn.Conv2d):
+    """nn.Conv2d whose forward goes through fda_conv2d (bias-free, as the
+    reference's Flux convs are — /root/reference README.md model usage)."""
+
+    def forward(self, x):
+        return fda_conv2d(x, self.weight, self.stride, self.padding,
+                          self.dilation, self.groups)
