@@ -607,6 +607,376 @@ void conv_skcombine_launch(const float* part, void* y, float* stats, long M,
                        (const unsigned short*)accsrc);
 }
 
+
+// ---- 8-phase fine-interleave variant (guide §5 "256² 8-phase template",
+// adapted to the conv gather): 256x128 tile, 8 waves (4M x 2N), 512
+// threads, 1 block/CU, 3 LDS buffers (144 KiB). Each K-step runs as 4
+// phases of [ds-read subtile | stage chunk of tile t+2 | raw barrier |
+// lgkmcnt(0) | 8 MFMAs | raw barrier]; the tile drain is ONE counted
+// s_waitcnt vmcnt(12) per K-step (2 newer tiles x 6 glds/wave stay in
+// flight), so DMA latency spans ~4-8 phases of MFMA work instead of
+// stalling at a single per-step barrier. FLUXDIST_CONV8 gates dispatch.
+template <int MODE>
+__global__ __launch_bounds__(512, 1) void conv_igemm8_kernel(
+    const unsigned short* __restrict__ src,
+    const unsigned short* __restrict__ wgt,
+    unsigned short* __restrict__ out,
+    int N, int H, int W, int C,
+    int K, int P, int Q,
+    int R, int S, int sy, int sx, int py, int px,
+    float* __restrict__ stats,
+    const unsigned short* __restrict__ accsrc) {
+    constexpr int BM = 256, BN = 128, NBUF = 3;
+    constexpr int A_ELEMS = BM * BK;
+    constexpr int B_ELEMS = BN * BK;
+    constexpr int BUF_ELEMS = A_ELEMS + B_ELEMS;
+    constexpr int AI = 4, BI = 2;        // glds per wave per tile
+    constexpr int NW = 8, WN = 2;
+
+    const int OC = (MODE == CONV_DGRAD) ? C : K;
+    const int RC = (MODE == CONV_FWD) ? C : K;
+
+    int a = 0, b = 0, OH, OW, r0 = 0, s0 = 0, nR = R, nS = S;
+    if (MODE != CONV_DGRAD) {
+        OH = P; OW = Q;
+    } else {
+        a = blockIdx.z / sx;  b = blockIdx.z % sx;
+        OH = (H - a + sy - 1) / sy;
+        OW = (W - b + sx - 1) / sx;
+        r0 = (a + py) % sy;  nR = (R - r0 + sy - 1) / sy;
+        s0 = (b + px) % sx;  nS = (S - s0 + sx - 1) / sx;
+        if (OH <= 0 || OW <= 0) return;
+        if (nR < 0) nR = 0;
+        if (nS < 0) nS = 0;
+    }
+    const long M = (long)N * OH * OW;
+    const long m0 = (long)blockIdx.x * BM;
+    if (m0 >= M) return;
+    const int n0 = blockIdx.y * BN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int wm = wid >> 1;
+    const int wn = wid & 1;
+
+    extern __shared__ unsigned short lds[];   // [NBUF][BUF_ELEMS]
+
+    // ---- per-lane staging descriptors (same gather as conv_igemm_kernel)
+    int a_row[AI];
+    long a_pix[AI];
+    int a_hb[AI], a_wb[AI];
+    bool a_mok[AI];
+    const int cslot = lane & 7;
+    #pragma unroll
+    for (int i = 0; i < AI; ++i) {
+        const int row = (wid * AI + i) * 8 + (lane >> 3);
+        a_row[i] = row;
+        const long m = m0 + row;
+        const bool mok = m < M;
+        const long mm = mok ? m : 0;
+        const int ow = (int)(mm % OW);
+        const int oh = (int)((mm / OW) % OH);
+        const int n = (int)(mm / ((long)OW * OH));
+        a_mok[i] = mok;
+        if (MODE == CONV_FWD) {
+            a_hb[i] = oh * sy - py;
+            a_wb[i] = ow * sx - px;
+            a_pix[i] = ((long)n * H) * W * C;
+        } else {
+            a_hb[i] = oh + (a + py) / sy;
+            a_wb[i] = ow + (b + px) / sx;
+            a_pix[i] = ((long)n * P) * Q * K;
+        }
+    }
+    int b_row[BI];
+    #pragma unroll
+    for (int i = 0; i < BI; ++i) b_row[i] = (wid * BI + i) * 8 + (lane >> 3);
+
+    const int cblocks = RC / BK;
+    const int T = nR * nS * cblocks;
+
+    long a_goff[AI];
+    bool a_okc[AI];
+    long b_goff[BI];
+    int last_rsi = -1;
+    auto advance = [&](int it) {
+        const int rsi = it / cblocks;
+        const int ri = rsi / nS, si = rsi % nS;
+        if (rsi != last_rsi) {
+            last_rsi = rsi;
+            const int cb = (it % cblocks) * BK;
+            #pragma unroll
+            for (int i = 0; i < AI; ++i) {
+                bool ok = a_mok[i];
+                long off = 0;
+                if (MODE == CONV_FWD) {
+                    const int h = a_hb[i] + ri, w = a_wb[i] + si;
+                    ok = ok && (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W;
+                    off = a_pix[i] + ((long)h * W + w) * C + cb + (cslot ^ (a_row[i] & 7)) * 8;
+                } else {
+                    const int pp = a_hb[i] - ri, qq = a_wb[i] - si;
+                    ok = ok && (unsigned)pp < (unsigned)P && (unsigned)qq < (unsigned)Q;
+                    off = a_pix[i] + ((long)pp * Q + qq) * K + cb + (cslot ^ (a_row[i] & 7)) * 8;
+                }
+                a_goff[i] = off;
+                a_okc[i] = ok;
+            }
+            const int r = r0 + ri * ((MODE == CONV_FWD) ? 1 : sy);
+            const int sss = s0 + si * ((MODE == CONV_FWD) ? 1 : sx);
+            const int rs = r * S + sss;
+            #pragma unroll
+            for (int i = 0; i < BI; ++i) {
+                const int row = b_row[i];
+                const int cs = (cslot ^ (row & 7)) * 8;
+                if (MODE == CONV_FWD)
+                    b_goff[i] = (long)(n0 + row) * R * S * C + (long)rs * C + cb + cs;
+                else
+                    b_goff[i] = (long)((long)rs * C + n0 + row) * K + cb + cs;
+            }
+        } else {
+            #pragma unroll
+            for (int i = 0; i < AI; ++i) a_goff[i] += BK;
+            #pragma unroll
+            for (int i = 0; i < BI; ++i) b_goff[i] += BK;
+        }
+    };
+
+    // fragment read offsets
+    int a_off[4][2], b_off[4][2];
+    {
+        const int fr = lane & 15, fq = lane >> 4;
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+            #pragma unroll
+            for (int kh = 0; kh < 2; ++kh) {
+                const int row = wm * 64 + mi * 16 + fr;
+                const int slot = (kh * 4 + fq) ^ (row & 7);
+                a_off[mi][kh] = row * BK + slot * 8;
+            }
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+            #pragma unroll
+            for (int kh = 0; kh < 2; ++kh) {
+                const int row = wn * 64 + ni * 16 + fr;
+                const int slot = (kh * 4 + fq) ^ (row & 7);
+                b_off[ni][kh] = A_ELEMS + row * BK + slot * 8;
+            }
+    }
+
+    floatx4 acc[4][4];
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) acc[mi][ni] = floatx4{0.f, 0.f, 0.f, 0.f};
+
+    auto issueA = [&](int buf, int i) {
+        unsigned short* base = lds + buf * BUF_ELEMS;
+        const unsigned short* sp = a_okc[i] ? src + a_goff[i] : conv_zero16;
+        FDA_GLDS16(sp, base + (wid * AI + i) * 8 * BK);
+    };
+    auto issueB = [&](int buf, int i) {
+        unsigned short* base = lds + buf * BUF_ELEMS;
+        FDA_GLDS16(wgt + b_goff[i], base + A_ELEMS + (wid * BI + i) * 8 * BK);
+    };
+    auto stage_all = [&](int buf, int it) {   // prologue only
+        advance(it);
+        #pragma unroll
+        for (int i = 0; i < AI; ++i) issueA(buf, i);
+        #pragma unroll
+        for (int i = 0; i < BI; ++i) issueB(buf, i);
+    };
+
+    if (T > 0) stage_all(0, 0);
+    if (T > 1) stage_all(1, 1);
+
+    for (int it = 0; it < T; ++it) {
+        const int buf = it % NBUF;
+        const unsigned short* rbuf = lds + buf * BUF_ELEMS;
+        const int sbuf = (it + 2) % NBUF;
+        const bool do_stage = it + 2 < T;
+        if (do_stage) advance(it + 2);
+        // drain THIS tile's 6 glds; leave the 2 newer tiles (<=12) in flight
+        if (it + 1 < T)
+            asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        short8 af0[4], af1[4], bf[4];
+        // ---- phase 0: kh=0, ni 0..1 ------------------------------------
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+            af0[mi] = *(const short8*)(rbuf + a_off[mi][0]);
+        bf[0] = *(const short8*)(rbuf + b_off[0][0]);
+        bf[1] = *(const short8*)(rbuf + b_off[1][0]);
+        if (do_stage) { issueA(sbuf, 0); issueA(sbuf, 1); }
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+            acc[mi][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af0[mi], bf[0], acc[mi][0], 0, 0, 0);
+            acc[mi][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af0[mi], bf[1], acc[mi][1], 0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();
+        // ---- phase 1: kh=0, ni 2..3 ------------------------------------
+        bf[2] = *(const short8*)(rbuf + b_off[2][0]);
+        bf[3] = *(const short8*)(rbuf + b_off[3][0]);
+        if (do_stage) { issueA(sbuf, 2); issueA(sbuf, 3); }
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+            acc[mi][2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af0[mi], bf[2], acc[mi][2], 0, 0, 0);
+            acc[mi][3] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af0[mi], bf[3], acc[mi][3], 0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();
+        // ---- phase 2: kh=1, ni 0..1 ------------------------------------
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+            af1[mi] = *(const short8*)(rbuf + a_off[mi][1]);
+        bf[0] = *(const short8*)(rbuf + b_off[0][1]);
+        bf[1] = *(const short8*)(rbuf + b_off[1][1]);
+        if (do_stage) issueB(sbuf, 0);
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+            acc[mi][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af1[mi], bf[0], acc[mi][0], 0, 0, 0);
+            acc[mi][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af1[mi], bf[1], acc[mi][1], 0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();
+        // ---- phase 3: kh=1, ni 2..3 ------------------------------------
+        bf[2] = *(const short8*)(rbuf + b_off[2][1]);
+        bf[3] = *(const short8*)(rbuf + b_off[3][1]);
+        if (do_stage) issueB(sbuf, 1);
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+            acc[mi][2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af1[mi], bf[2], acc[mi][2], 0, 0, 0);
+            acc[mi][3] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af1[mi], bf[3], acc[mi][3], 0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();
+    }
+
+    // ---- epilogue (same mapping as conv_igemm_kernel) --------------------
+    const int fcol = lane & 15, frow0 = (lane >> 4) * 4;
+    float ssum[4] = {0.f, 0.f, 0.f, 0.f};
+    float sq[4] = {0.f, 0.f, 0.f, 0.f};
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const long m = m0 + wm * 64 + mi * 16 + frow0 + j;
+            if (m >= M) continue;
+            long obase;
+            if (MODE != CONV_DGRAD) {
+                obase = m * OC;
+            } else {
+                const int ww = (int)(m % OW);
+                const int hh = (int)((m / OW) % OH);
+                const int n = (int)(m / ((long)OW * OH));
+                obase = (((long)n * H + a + (long)sy * hh) * W + b +
+                         (long)sx * ww) * C;
+            }
+            unsigned short* orow = out + obase + n0 + wn * 64;
+            const unsigned short* arow =
+                accsrc ? accsrc + obase + n0 + wn * 64 : nullptr;
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni) {
+                float v = acc[mi][ni][j];
+                if (arow) v += bf16bits_to_f32(arow[ni * 16 + fcol]);
+                const unsigned short us = f32_to_bf16bits(v);
+                orow[ni * 16 + fcol] = us;
+                if (MODE != CONV_DGRAD && stats != nullptr) {
+                    const float vr = bf16bits_to_f32(us);
+                    ssum[ni] += vr;
+                    sq[ni] += vr * vr;
+                }
+            }
+        }
+    }
+    if (MODE != CONV_DGRAD && stats != nullptr) {
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+            #pragma unroll
+            for (int off = 16; off < 64; off <<= 1) {
+                ssum[ni] += __shfl_xor(ssum[ni], off, 64);
+                sq[ni] += __shfl_xor(sq[ni], off, 64);
+            }
+        }
+        float* sf = (float*)lds;
+        __builtin_amdgcn_s_barrier();
+        if ((lane >> 4) == 0) {
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni) {
+                sf[(wid * 4 + ni) * 16 + fcol] = ssum[ni];
+                sf[NW * 64 + (wid * 4 + ni) * 16 + fcol] = sq[ni];
+            }
+        }
+        __builtin_amdgcn_s_barrier();
+        if (wm == 0 && (lane >> 4) == 0) {
+            float* prow = stats + (long)blockIdx.x * 2 * OC + n0 + wn * 64;
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni) {
+                float s2 = ssum[ni], z = sq[ni];
+                for (int w2 = wid + WN; w2 < NW; w2 += WN) {
+                    s2 += sf[(w2 * 4 + ni) * 16 + fcol];
+                    z += sf[NW * 64 + (w2 * 4 + ni) * 16 + fcol];
+                }
+                prow[ni * 16 + fcol] = s2;
+                prow[OC + ni * 16 + fcol] = z;
+            }
+        }
+    }
+}
+
+int conv8_enabled() {
+    // 8-phase variant A/B knob: FLUXDIST_CONV8=1 routes eligible shapes
+    // (OC%128, SK==1, grid fills at 1 block/CU) through conv_igemm8_kernel.
+    static int v = [] {
+        const char* e = getenv("FLUXDIST_CONV8");
+        return e ? atoi(e) : 0;
+    }();
+    return v;
+}
+
+template <int MODE>
+static void launch_cfg8(const void* src, const void* wgt, void* out,
+                        int N, int H, int W, int C, int K, int P, int Q,
+                        int R, int S, int sy, int sx, int py, int px,
+                        hipStream_t stream, float* stats,
+                        const void* accsrc) {
+    const int OC = (MODE == CONV_DGRAD) ? C : K;
+    const long M = (MODE != CONV_DGRAD)
+        ? (long)N * P * Q
+        : (long)N * ((H + sy - 1) / sy) * ((W + sx - 1) / sx);
+    const unsigned zbase = (MODE != CONV_DGRAD) ? 1u : (unsigned)(sy * sx);
+    dim3 grid((unsigned)((M + 255) / 256), (unsigned)(OC / 128), zbase);
+    const size_t shmem = 3 * (256 * BK + 128 * BK) * sizeof(unsigned short);
+    static bool raised = [] {
+        hipFuncSetAttribute((const void*)&conv_igemm8_kernel<MODE>,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
+        return true;
+    }();
+    (void)raised;
+    hipLaunchKernelGGL((conv_igemm8_kernel<MODE>), grid, dim3(512), shmem,
+                       stream, (const unsigned short*)src,
+                       (const unsigned short*)wgt, (unsigned short*)out,
+                       N, H, W, C, K, P, Q, R, S, sy, sx, py, px, stats,
+                       (const unsigned short*)accsrc);
+}
+
 void conv_stem_fwd_launch(const void* src, const void* wgt, void* out,
                           int N, int Hp, int Wp, int K, int P, int Q,
                           int R, int sy, int sx, hipStream_t stream,
