@@ -1012,6 +1012,8 @@ static int conv_bigtile() {
     return v;
 }
 
+int conv8_enabled();   // defined below (8-phase variant knob)
+
 bool conv_use_inlsk() {
     // in-launch split-K combine (last-arriver seam) vs the separate
     // combine kernel. Default OFF: measured 2-2.6x SLOWER than the
@@ -1052,7 +1054,7 @@ void conv_igemm_plan(long M, int OC, long T, int zbase,
         if (blocks < 192) SK = 4;
         else if (blocks < 384) SK = 2;
     }
-    if (big && SK == 1 && conv_bigtile()) {
+    if (big && SK == 1 && (conv_bigtile() || conv8_enabled())) {
         const long blocks256 = ((M + 255) / 256) * (OC / 128) * zbase;
         if (blocks256 >= 256) { BM = 256; BN = 128; }
     }
@@ -1078,6 +1080,17 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
     conv_igemm_plan(Mv, OC, (long)R * S * (RC / 64), zbase, &BM, &BN, &SKp);
     const bool big = BN == 128;
     if (BM == 256 && BN == 128) {
+        if (conv8_enabled()) {     // 8-phase fine-interleave variant
+            if (dgrad)
+                launch_cfg8<CONV_DGRAD>(src, wgt, out, N, H, W, C, K, P, Q,
+                                        R, S, sy, sx, py, px, stream,
+                                        nullptr, accsrc);
+            else
+                launch_cfg8<CONV_FWD>(src, wgt, out, N, H, W, C, K, P, Q,
+                                      R, S, sy, sx, py, px, stream, stats,
+                                      accsrc);
+            return;
+        }
         if (dgrad)
             launch_cfg<CONV_DGRAD, 256, 128, 2, 2>(
                 src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
@@ -1085,7 +1098,7 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
         else
             launch_cfg<CONV_FWD, 256, 128, 2, 2>(
                 src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
-                stream, stats, skpart, SK, cnt);
+                stream, stats, skpart, SK, cnt, accsrc);
         return;
     }
     if (dgrad) {
