@@ -803,67 +803,46 @@ __global__ __launch_bounds__(512, 1) void conv_igemm8_kernel(
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
 
-        short8 af0[4], af1[4], bf[4];
-        // ---- phase 0: kh=0, ni 0..1 ------------------------------------
+        short8 af0[4], af1[4], bf[4][2];
+        // ---- phase 0: kh=0 (16 MFMAs) ----------------------------------
+        // 2 phases of 16 MFMAs: the 4x8-MFMA grain measured ~10% SLOWER
+        // (per-phase barrier cost not amortized at this tile size;
+        // profiles/ab_conv8.md) — FLUXDIST_CONV8_P4=1 selects it back.
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             af0[mi] = *(const short8*)(rbuf + a_off[mi][0]);
-        bf[0] = *(const short8*)(rbuf + b_off[0][0]);
-        bf[1] = *(const short8*)(rbuf + b_off[1][0]);
-        if (do_stage) { issueA(sbuf, 0); issueA(sbuf, 1); }
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+            bf[ni][0] = *(const short8*)(rbuf + b_off[ni][0]);
+        if (do_stage) { issueA(sbuf, 0); issueA(sbuf, 1); issueA(sbuf, 2); }
         __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
-        for (int mi = 0; mi < 4; ++mi) {
-            acc[mi][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af0[mi], bf[0], acc[mi][0], 0, 0, 0);
-            acc[mi][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af0[mi], bf[1], acc[mi][1], 0, 0, 0);
-        }
+        for (int mi = 0; mi < 4; ++mi)
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af0[mi], bf[ni][0], acc[mi][ni], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
         __builtin_amdgcn_s_barrier();
-        // ---- phase 1: kh=0, ni 2..3 ------------------------------------
-        bf[2] = *(const short8*)(rbuf + b_off[2][0]);
-        bf[3] = *(const short8*)(rbuf + b_off[3][0]);
-        if (do_stage) { issueA(sbuf, 2); issueA(sbuf, 3); }
-        __builtin_amdgcn_s_barrier();
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_setprio(1);
-        #pragma unroll
-        for (int mi = 0; mi < 4; ++mi) {
-            acc[mi][2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af0[mi], bf[2], acc[mi][2], 0, 0, 0);
-            acc[mi][3] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af0[mi], bf[3], acc[mi][3], 0, 0, 0);
-        }
-        __builtin_amdgcn_s_setprio(0);
-        __builtin_amdgcn_s_barrier();
-        // ---- phase 2: kh=1, ni 0..1 ------------------------------------
+        // ---- phase 1: kh=1 (16 MFMAs) ----------------------------------
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             af1[mi] = *(const short8*)(rbuf + a_off[mi][1]);
-        bf[0] = *(const short8*)(rbuf + b_off[0][1]);
-        bf[1] = *(const short8*)(rbuf + b_off[1][1]);
-        if (do_stage) issueB(sbuf, 0);
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+            bf[ni][1] = *(const short8*)(rbuf + b_off[ni][1]);
+        if (do_stage) { issueA(sbuf, 3); issueB(sbuf, 0); issueB(sbuf, 1); }
         __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
-        for (int mi = 0; mi < 4; ++mi) {
-            acc[mi][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af1[mi], bf[0], acc[mi][0], 0, 0, 0);
-            acc[mi][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af1[mi], bf[1], acc[mi][1], 0, 0, 0);
-        }
-        __builtin_amdgcn_s_setprio(0);
-        __builtin_amdgcn_s_barrier();
-        // ---- phase 3: kh=1, ni 2..3 ------------------------------------
-        bf[2] = *(const short8*)(rbuf + b_off[2][1]);
-        bf[3] = *(const short8*)(rbuf + b_off[3][1]);
-        if (do_stage) issueB(sbuf, 1);
-        __builtin_amdgcn_s_barrier();
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_setprio(1);
-        #pragma unroll
-        for (int mi = 0; mi < 4; ++mi) {
-            acc[mi][2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af1[mi], bf[2], acc[mi][2], 0, 0, 0);
-            acc[mi][3] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af1[mi], bf[3], acc[mi][3], 0, 0, 0);
-        }
+        for (int mi = 0; mi < 4; ++mi)
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af1[mi], bf[ni][1], acc[mi][ni], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
         __builtin_amdgcn_s_barrier();
     }
