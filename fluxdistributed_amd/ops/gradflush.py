@@ -27,10 +27,37 @@ def _bucketer_active() -> bool:
     return bool(_NOTIFY)
 
 
+# Replica threads flag their backwards as non-deferrable (task_ddp sets
+# this around train_step); autograd device workers inherit nothing from
+# the launching thread, so the flag lives in a process-global counter.
+_NO_DEFER = [0]
+
+
+def push_no_defer():
+    _NO_DEFER[0] += 1
+
+
+def pop_no_defer():
+    _NO_DEFER[0] -= 1
+
+
+def _deferral_ok() -> bool:
+    return _NO_DEFER[0] == 0
+
+
 def queue_or_flush(param, g_sl: torch.Tensor, ws_flat: torch.Tensor) -> None:
-    """G slice += cast(ws): now (DDP) or batched at backward end."""
+    """G slice += cast(ws): now (DDP / replica threads) or batched at
+    backward end (single main-thread training).
+
+    The deferred path is gated to backwards whose GRAPHS were launched
+    from the main thread: with several task-DDP replica threads running
+    concurrent backwards, all device nodes execute interleaved on ONE
+    autograd device worker, and one graph's end-callback would flush the
+    other graph's still-accumulating queue entries — correct but
+    unverifiable ordering; replicas keep the per-layer immediate flush.
+    """
     C = require_native("grad_accum")
-    if _bucketer_active():
+    if _bucketer_active() or not _deferral_ok():
         from ..parallel.bucketing import notify_grad_written
 
         C.grad_accum_bf16(g_sl, ws_flat)

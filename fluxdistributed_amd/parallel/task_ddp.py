@@ -118,11 +118,19 @@ def prepare_training(
 def train_step(loss_fn, buffer, replica: Replica, x, y):
     """Forward+backward on one replica, then publish grads to its buffer slot
     (/root/reference/src/ddp_tasks.jl:80-84)."""
+    from ..ops.gradflush import pop_no_defer, push_no_defer
+
     with device_ctx(replica.device):
         replica.optimizer.zero_grad()
         out = replica.model(x)
         loss = loss_fn(out, y)
-        loss.backward()
+        # concurrent replica backwards share the autograd device worker:
+        # direct-grad flushes must stay per-layer immediate (gradflush.py)
+        push_no_defer()
+        try:
+            loss.backward()
+        finally:
+            pop_no_defer()
         markbuffer_(buffer[replica.index], grads_of(replica.model))
         synchronize(replica.device)
     return loss.detach()
