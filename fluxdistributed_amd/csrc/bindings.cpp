@@ -253,6 +253,9 @@ static unsigned* sk_tickets(const at::Tensor& ref, long ntiles,
 at::Tensor conv_igemm_fwd(at::Tensor x, at::Tensor w,
                           int64_t sy, int64_t sx, int64_t py, int64_t px) {
     TORCH_CHECK(x.dim() == 4 && w.dim() == 4);
+    TORCH_CHECK(x.is_cuda() && w.is_cuda(),
+                "conv_igemm: device tensors required (a CPU tensor here "
+                "would fault the GPU with a host pointer)");
     TORCH_CHECK(x.scalar_type() == at::kBFloat16 && w.scalar_type() == at::kBFloat16,
                 "conv_igemm: bf16 only");
     TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
@@ -368,6 +371,7 @@ at::Tensor conv_igemm_dgrad(at::Tensor dy, at::Tensor wt,
     // the epilogue (residual-junction grad fusion, ops/conv.py).
     // dy: [N,K,P,Q] channels_last; wt: [R*S*C, K] row-major (pre-transposed
     // weight, k contiguous). Output dx: [N,C,H,W] channels_last.
+    TORCH_CHECK(dy.is_cuda() && wt.is_cuda());
     TORCH_CHECK(dy.dim() == 4 && dy.scalar_type() == at::kBFloat16);
     TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast));
     TORCH_CHECK(wt.dim() == 2 && wt.is_contiguous() &&
@@ -430,6 +434,7 @@ at::Tensor conv_igemm_wgrad(at::Tensor dy, at::Tensor x,
     // dy: [N,K,P,Q] channels_last bf16; x: [N,C,H,W] channels_last bf16.
     // Returns ws[K][R*S*C] fp32 (the channels_last weight-grad memory
     // layout [K][R][S][C] flattened).
+    TORCH_CHECK(dy.is_cuda() && x.is_cuda());
     TORCH_CHECK(dy.scalar_type() == at::kBFloat16 &&
                 x.scalar_type() == at::kBFloat16);
     TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) &&
@@ -470,6 +475,7 @@ at::Tensor conv_stem_fwd(at::Tensor x8, at::Tensor wpad, int64_t R,
                          int64_t sy, int64_t sx, int64_t P, int64_t Q) {
     // x8: [N,8,Hp,Wp] channels_last bf16 (spatially pre-padded, channels
     // 3..7 zero); wpad: [K][R][64] bf16 (s==7 and c>=3 taps zero).
+    TORCH_CHECK(x8.is_cuda() && wpad.is_cuda());
     TORCH_CHECK(x8.scalar_type() == at::kBFloat16 &&
                 x8.is_contiguous(at::MemoryFormat::ChannelsLast));
     TORCH_CHECK(x8.size(1) == 8);

@@ -12,7 +12,12 @@ import torch
 
 
 def is_real_gpu(dev) -> bool:
-    return isinstance(dev, torch.device) and dev.type == "cuda"
+    # accepts torch.device OR its string form ("cuda", "cuda:0"): a string
+    # device silently classifying as logical left replicas on the CPU and
+    # fed host pointers to device kernels (GPU fault, r2)
+    if isinstance(dev, torch.device):
+        return dev.type == "cuda"
+    return isinstance(dev, str) and dev.startswith("cuda")
 
 
 @contextlib.contextmanager
