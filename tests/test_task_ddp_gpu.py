@@ -48,27 +48,30 @@ def test_bf16_logical_fanout_on_one_gpu():
     for k in p0:
         assert torch.equal(p0[k].float(), p1[k].float()), f"divergence in {k}"
 
-    # solo large-batch oracle from the same init
+    # Sequential shard-average oracle — the SAME algorithm single-threaded.
+    # NOT a large-batch oracle: BatchNorm stats are per-replica over each
+    # 4-sample shard, so large-batch equivalence doesn't hold in train
+    # mode (the reference tests under testmode! for exactly this reason,
+    # SURVEY.md §4). Here: per-shard grads on one model, averaged, one
+    # step — mathematically identical to the fan-out, to bf16 reorder.
     solo = _model().to("cuda:0")
     opt = FusedSGDMomentum(solo.parameters(), lr=0.05, momentum=0.9)
     for step in range(3):
-        (x0, y0), (x1, y1) = _shards(step)
-        x = torch.cat([x0, x1]).contiguous(memory_format=torch.channels_last)
-        y = torch.cat([y0, y1])
-        opt.zero_grad()
-        logit_cross_entropy(solo(x), y).backward()
+        shard_grads = []
+        for (x, y) in _shards(step):
+            opt.zero_grad()
+            logit_cross_entropy(solo(x), y).backward()
+            torch.cuda.synchronize()
+            shard_grads.append([g.G.float().clone() for g in opt.groups])
+        with torch.no_grad():
+            for g, g0, g1 in zip(opt.groups, *shard_grads):
+                g.G.copy_(((g0 + g1) / 2).to(g.G.dtype))
         opt.step()
     torch.cuda.synchronize()
     ps = dict(solo.named_parameters())
-    # bf16 fan-out (grads averaged in bf16 over the HOST buffer) vs a solo
-    # large batch diverges per step by bf16 rounding; after 3 momentum
-    # steps the max-element drift on slow-moving early weights measured
-    # ~0.05 on a 0.2-max-weight stem, so the per-element bar is loose and
-    # a tight bar is put on the RELATIVE drift of the whole tensor.
     for k in p0:
         a, b = p0[k].float(), ps[k].float()
         scale = float(b.abs().max())
         err = float((a - b).abs().max())
-        rel = float((a - b).norm() / (b.norm() + 1e-8))
-        assert err < 0.30 * max(scale, 1e-2) and rel < 0.05, \
-            f"{k}: drift {err} scale {scale} rel {rel}"
+        assert err < 0.02 * max(scale, 1e-2), \
+            f"{k}: drift {err} scale {scale}"
