@@ -1195,7 +1195,7 @@ typedef __attribute__((ext_vector_type(4))) short short4_;
 constexpr int WG_BM = 64;      // m per K-step
 constexpr int WG_MCH = 2048;   // pixels per block (chunk)
 
-template <int FT, bool STEM = false>
+template <int FT, bool STEM = false, bool ASMRD = false>
 __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
     const unsigned short* __restrict__ dy,   // [M][K] (NHWC out grad)
     const unsigned short* __restrict__ x,    // [N,H,W,C]
@@ -1333,6 +1333,90 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
         // DMA now flies over the MFMA cluster.
         const unsigned short* buf = lds + (it % RING) * 2 * TILE_ELEMS;
         short4_ a[FT][2][2], b[FT][2][2];   // [fi][ks][half]
+        if constexpr (ASMRD) {
+            // tr16 reads via inline asm, INVISIBLE to hipcc's wait
+            // inserter: the compiler otherwise (a) force-drains all
+            // outstanding LDS-DMA with vmcnt(0) before every visible
+            // tr-read cluster and (b) waits lgkmcnt(0) over ALL 8*FT
+            // reads before the first MFMA. Here the waits are counted by
+            // hand: ks=1's reads stay in flight under ks=0's MFMAs.
+            // stage() goes FIRST (max DMA flight time): its stray
+            // compiler lgkmcnt(0) then fires while nothing is
+            // outstanding, instead of draining the tr reads.
+            if (it + RING - 1 < nsteps) stage((it + RING - 1) % RING);
+            __builtin_amdgcn_sched_barrier(0);
+            #pragma unroll
+            for (int ks = 0; ks < 2; ++ks)
+                #pragma unroll
+                for (int fi = 0; fi < FT; ++fi) {
+                    const int kb_a = wk * FT + fi;
+                    const int kb_b = wc * FT + fi;
+                    const int v0 = ks * 8 + lg * 2;
+                    const int p0 = (v0 >> 1);
+                    const int p1 = p0 + 8;
+                    const unsigned short* pa = buf + kb_a * 1024 + l15 * 4;
+                    const unsigned short* pb =
+                        buf + TILE_ELEMS + kb_b * 1024 + l15 * 4;
+                    const unsigned aa0 = (unsigned)(unsigned long)
+                        (const __attribute__((address_space(3))) char*)
+                        (const char*)(pa + p0 * 64);
+                    const unsigned aa1 = (unsigned)(unsigned long)
+                        (const __attribute__((address_space(3))) char*)
+                        (const char*)(pa + p1 * 64);
+                    const unsigned ab0 = (unsigned)(unsigned long)
+                        (const __attribute__((address_space(3))) char*)
+                        (const char*)(pb + p0 * 64);
+                    const unsigned ab1 = (unsigned)(unsigned long)
+                        (const __attribute__((address_space(3))) char*)
+                        (const char*)(pb + p1 * 64);
+                    asm volatile("ds_read_b64_tr_b16 %0, %1"
+                                 : "=v"(a[fi][ks][0]) : "v"(aa0));
+                    asm volatile("ds_read_b64_tr_b16 %0, %1"
+                                 : "=v"(a[fi][ks][1]) : "v"(aa1));
+                    asm volatile("ds_read_b64_tr_b16 %0, %1"
+                                 : "=v"(b[fi][ks][0]) : "v"(ab0));
+                    asm volatile("ds_read_b64_tr_b16 %0, %1"
+                                 : "=v"(b[fi][ks][1]) : "v"(ab1));
+                }
+            __builtin_amdgcn_s_setprio(1);
+            #pragma unroll
+            for (int ks = 0; ks < 2; ++ks) {
+                // sched_barrier(0) pins the MFMA clusters BETWEEN the
+                // counted waits: MFMAs are pure-register ops, so without
+                // it LLVM floats both clusters past both waits — and the
+                // asm-read results carry no wait-dependency at all (the
+                // hand-counted waits ARE the only correctness fence).
+                __builtin_amdgcn_sched_barrier(0);
+                if (ks == 0) {
+                    // lgkmcnt is 4-bit (max 15): for FT=4 the ideal
+                    // "leave 16 in flight" clamps to 15 (ds_reads return
+                    // in order, so this waits one extra ks=1 read)
+                    if constexpr (FT == 4)
+                        asm volatile("s_waitcnt lgkmcnt(15)" ::: "memory");
+                    else
+                        asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+                } else {
+                    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+                }
+                __builtin_amdgcn_sched_barrier(0);
+                #pragma unroll
+                for (int ki = 0; ki < FT; ++ki)
+                    #pragma unroll
+                    for (int ci = 0; ci < FT; ++ci) {
+                        short8 af, bf;
+                        #pragma unroll
+                        for (int e = 0; e < 4; ++e) {
+                            af[e] = a[ki][ks][0][e];
+                            af[e + 4] = a[ki][ks][1][e];
+                            bf[e] = b[ci][ks][0][e];
+                            bf[e + 4] = b[ci][ks][1][e];
+                        }
+                        acc[ki][ci] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            af, bf, acc[ki][ci], 0, 0, 0);
+                    }
+            }
+            __builtin_amdgcn_s_setprio(0);
+        } else {
         #pragma unroll
         for (int fi = 0; fi < FT; ++fi)
             #pragma unroll
@@ -1374,6 +1458,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
                         af, bf, acc[ki][ci], 0, 0, 0);
                 }
         __builtin_amdgcn_s_setprio(0);
+        }
         // end barrier dropped: the next iteration's wait + top barrier
         // orders buffer reuse for every ring depth.
     }
@@ -1394,6 +1479,16 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
             }
 }
 
+static bool wgrad_asm() {
+    // A/B knob: hand-counted lgkm waits via inline-asm tr16 reads
+    // (FLUXDIST_WGRAD_ASM=0 restores the intrinsic path).
+    static bool v = [] {
+        const char* e = getenv("FLUXDIST_WGRAD_ASM");
+        return !(e && e[0] == '0');
+    }();
+    return v;
+}
+
 void conv_stem_wgrad_launch(const void* dy, const void* x, float* ws,
                             int N, int Hp, int Wp, int K, int P, int Q,
                             int R, int sy, int sx, hipStream_t stream) {
@@ -1406,10 +1501,17 @@ void conv_stem_wgrad_launch(const void* dy, const void* x, float* ws,
     const int nch = (int)((M + STEM_MCH - 1) / STEM_MCH);
     dim3 grid((unsigned)(K / 64), 1u, (unsigned)(R * nch));
     const size_t shmem = 3 * 2 * (WG_BM * 64) * sizeof(unsigned short);
-    hipLaunchKernelGGL((conv_wgrad_kernel<2, true>), grid, dim3(256), shmem,
-                       stream, (const unsigned short*)dy,
-                       (const unsigned short*)x, ws, N, Hp, Wp, /*C=*/64, K,
-                       P, Q, R, /*S=*/1, sy, sx, 0, 0, nch, STEM_MCH);
+    if (wgrad_asm())
+        hipLaunchKernelGGL((conv_wgrad_kernel<2, true, true>), grid,
+                           dim3(256), shmem, stream,
+                           (const unsigned short*)dy,
+                           (const unsigned short*)x, ws, N, Hp, Wp, 64, K,
+                           P, Q, R, 1, sy, sx, 0, 0, nch, STEM_MCH);
+    else
+        hipLaunchKernelGGL((conv_wgrad_kernel<2, true>), grid, dim3(256),
+                           shmem, stream, (const unsigned short*)dy,
+                           (const unsigned short*)x, ws, N, Hp, Wp, 64, K,
+                           P, Q, R, 1, sy, sx, 0, 0, nch, STEM_MCH);
 }
 
 // pick the pixel-chunk size so the grid lands near `target` blocks:
@@ -1439,10 +1541,17 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
         dim3 grid((unsigned)(K / 128), (unsigned)(C / 128),
                   (unsigned)(R * S * nch));
         const size_t shmem = 2 * 2 * (WG_BM * 128) * sizeof(unsigned short);
-        hipLaunchKernelGGL(conv_wgrad_kernel<4>, grid, dim3(256), shmem,
-                           stream, (const unsigned short*)dy,
-                           (const unsigned short*)x, ws, N, H, W, C, K, P, Q,
-                           R, S, sy, sx, py, px, nch, mch);
+        if (wgrad_asm())
+            hipLaunchKernelGGL((conv_wgrad_kernel<4, false, true>), grid,
+                               dim3(256), shmem, stream,
+                               (const unsigned short*)dy,
+                               (const unsigned short*)x, ws, N, H, W, C, K,
+                               P, Q, R, S, sy, sx, py, px, nch, mch);
+        else
+            hipLaunchKernelGGL(conv_wgrad_kernel<4>, grid, dim3(256), shmem,
+                               stream, (const unsigned short*)dy,
+                               (const unsigned short*)x, ws, N, H, W, C, K,
+                               P, Q, R, S, sy, sx, py, px, nch, mch);
     } else {
         const long tiles2 = (long)(K / 64) * (C / 64) * R * S;
         const int mch = pick_mch(M, tiles2, 768);
@@ -1450,10 +1559,17 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
         dim3 grid((unsigned)(K / 64), (unsigned)(C / 64),
                   (unsigned)(R * S * nch));
         const size_t shmem = 3 * 2 * (WG_BM * 64) * sizeof(unsigned short);
-        hipLaunchKernelGGL(conv_wgrad_kernel<2>, grid, dim3(256), shmem,
-                           stream, (const unsigned short*)dy,
-                           (const unsigned short*)x, ws, N, H, W, C, K, P, Q,
-                           R, S, sy, sx, py, px, nch, mch);
+        if (wgrad_asm())
+            hipLaunchKernelGGL((conv_wgrad_kernel<2, false, true>), grid,
+                               dim3(256), shmem, stream,
+                               (const unsigned short*)dy,
+                               (const unsigned short*)x, ws, N, H, W, C, K,
+                               P, Q, R, S, sy, sx, py, px, nch, mch);
+        else
+            hipLaunchKernelGGL(conv_wgrad_kernel<2>, grid, dim3(256), shmem,
+                               stream, (const unsigned short*)dy,
+                               (const unsigned short*)x, ws, N, H, W, C, K,
+                               P, Q, R, S, sy, sx, py, px, nch, mch);
     }
 }
 
