@@ -1480,11 +1480,14 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
 }
 
 static bool wgrad_asm() {
-    // A/B knob: hand-counted lgkm waits via inline-asm tr16 reads
-    // (FLUXDIST_WGRAD_ASM=0 restores the intrinsic path).
+    // A/B knob FLUXDIST_WGRAD_ASM=1: hand-counted lgkm waits via
+    // inline-asm tr16 reads (ks-split MFMA overlap). Default OFF:
+    // measured -4.5% on the weighted wgrad total (2382 vs 2279 us) —
+    // the sched_barrier fences cost more scheduling freedom than the
+    // counted waits recover (profiles/ab_conv8.md campaign).
     static bool v = [] {
         const char* e = getenv("FLUXDIST_WGRAD_ASM");
-        return !(e && e[0] == '0');
+        return e && e[0] == '1';
     }();
     return v;
 }
