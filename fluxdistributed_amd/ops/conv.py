@@ -259,6 +259,10 @@ def take_conv_stats(x):
 
 
 def _want_conv_stats() -> bool:
+    """Decided in the DISPATCHER (fda_conv2d), where grad mode is visible —
+    inside Function.forward torch.is_grad_enabled() is always False, which
+    silently killed the conv->BN stats fusion (r2 profile: all 36 BNs ran
+    their own stats pass despite the handshake)."""
     return (torch.is_grad_enabled()
             and os.environ.get("FLUXDIST_BN_FUSE", "1") != "0")
 
@@ -281,14 +285,14 @@ class _FdaConv2d(torch.autograd.Function):
     via the library (aten convolution_backward with weight-only mask)."""
 
     @staticmethod
-    def forward(ctx, x, weight, stride, padding):
+    def forward(ctx, x, weight, stride, padding, want_stats):
         C = require_native("conv_igemm_fwd")
         clear_conv_stats()
         xc = x.contiguous(memory_format=torch.channels_last)
         wc = weight.contiguous(memory_format=torch.channels_last)
         sy, sx = stride
         py, px = padding
-        if _want_conv_stats():
+        if want_stats:
             y, part = C.conv_igemm_fwd_stats(xc, wc, sy, sx, py, px)
             stash_conv_stats(y, part)
         else:
@@ -353,7 +357,7 @@ class _FdaConv2d(torch.autograd.Function):
                 dw = torch.ops.aten.convolution_backward(
                     gy, x, w, None, list(stride), list(padding), [1, 1],
                     False, [0, 0], 1, [False, True, False])[1]
-        return dx, dw, None, None
+        return dx, dw, None, None, None
 
 
 class _FdaStemConv2d(torch.autograd.Function):
@@ -363,7 +367,7 @@ class _FdaStemConv2d(torch.autograd.Function):
     data); wgrad via conv_stem_wgrad."""
 
     @staticmethod
-    def forward(ctx, x, weight, stride, padding):
+    def forward(ctx, x, weight, stride, padding, want_stats):
         C = require_native("conv_stem_fwd")
         clear_conv_stats()
         K, Cin, R, S = weight.shape
@@ -391,7 +395,7 @@ class _FdaStemConv2d(torch.autograd.Function):
         wpad.view(K, R, 8, 8)[:, :, :S, :Cin] = (
             weight.contiguous(memory_format=torch.channels_last)
             .permute(0, 2, 3, 1))  # [K][R][S][C]
-        if _want_conv_stats():
+        if want_stats:
             y, part = C.conv_stem_fwd_stats(x8, wpad, R, sy, sx, P, Q)
             stash_conv_stats(y, part)
         else:
@@ -435,7 +439,7 @@ class _FdaStemConv2d(torch.autograd.Function):
                   .to(torch.bfloat16).permute(0, 3, 1, 2))
         ctx.x8 = None
         _stem_pool_put(ctx.pool_key, x8)
-        return None, dw, None, None
+        return None, dw, None, None, None
 
 
 def _stem_supported(x, weight, stride, padding, dilation, groups) -> bool:
@@ -458,9 +462,10 @@ def fda_conv2d(x: torch.Tensor, weight: torch.Tensor, stride=(1, 1),
     """conv2d with per-shape dispatch to the native MFMA kernel."""
     stride, padding, dilation = _pair(stride), _pair(padding), _pair(dilation)
     if _native_supported(x, weight, stride, padding, dilation, groups):
-        return _FdaConv2d.apply(x, weight, stride, padding)
+        return _FdaConv2d.apply(x, weight, stride, padding, _want_conv_stats())
     if _stem_supported(x, weight, stride, padding, dilation, groups):
-        return _FdaStemConv2d.apply(x, weight, stride, padding)
+        return _FdaStemConv2d.apply(x, weight, stride, padding,
+                                    _want_conv_stats())
     if os.environ.get("FLUXDIST_CONV", "") == "fda" and x.is_cuda:
         raise RuntimeError(
             f"FLUXDIST_CONV=fda but shape unsupported by conv_igemm: "

@@ -158,7 +158,7 @@ def test_stem_conv_fwd_and_wgrad(shape):
     xb = x.bfloat16().contiguous(memory_format=torch.channels_last)
     wb = wt.bfloat16().contiguous(memory_format=torch.channels_last).requires_grad_(True)
     assert _stem_supported(xb, wb, (s, s), (pad, pad), (1, 1), 1)
-    y = _FdaStemConv2d.apply(xb, wb, (s, s), (pad, pad))
+    y = _FdaStemConv2d.apply(xb, wb, (s, s), (pad, pad), False)
     err = (y.float() - ref).abs().max().item()
     scale = ref.abs().max().item()
     assert err < 0.03 * max(scale, 1.0), f"fwd err={err} scale={scale}"
