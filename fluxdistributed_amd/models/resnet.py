@@ -114,8 +114,12 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         if self.downsample is not None:
+            # downsample runs FIRST: python arg-evaluation order otherwise
+            # clobbers conv2's BN-stats stash with the downsample conv's
+            # (the 3 residual bn_stats launches in the r2 profile)
+            identity = self.downsample(x)
             out = self.bn1(self.conv1(x))
-            return self.bn2(self.conv2(out), residual=self.downsample(x))
+            return self.bn2(self.conv2(out), residual=identity)
         out = self.bn1(self.conv1(x))
         return self.bn2(self.conv2(out), residual=x,
                         defer_gres=_junction_fusible(x, self.conv1))
@@ -137,9 +141,10 @@ class Bottleneck(nn.Module):
 
     def forward(self, x):
         if self.downsample is not None:
+            identity = self.downsample(x)   # first — see BasicBlock note
             out = self.bn1(self.conv1(x))
             out = self.bn2(self.conv2(out))
-            return self.bn3(self.conv3(out), residual=self.downsample(x))
+            return self.bn3(self.conv3(out), residual=identity)
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
         return self.bn3(self.conv3(out), residual=x,
