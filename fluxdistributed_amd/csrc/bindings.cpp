@@ -101,8 +101,21 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_act_fwd(
         // stats already accumulated by the producing conv's epilogue
         auto& cp = *conv_part;
         TORCH_CHECK(cp.dim() == 3 && cp.size(1) == 2 && cp.size(2) == C);
+        const float* pp = cp.data_ptr<float>();
+        int NB = (int)cp.size(0);
+        at::Tensor p2;
+        if (NB > 1024) {
+            // layer-1-sized partial sets (~4700 m-tiles) fold at full
+            // grid width first; finalize then reads <=512 rows
+            const int NB2 = 512;
+            p2 = at::empty({NB2, 2, C}, fopts);
+            fda::bn_partial_prefold_launch(pp, p2.data_ptr<float>(), NB,
+                                           NB2, (int)C, stream);
+            pp = p2.data_ptr<float>();
+            NB = NB2;
+        }
         fda::bn_finalize_from_partials_launch(
-            cp.data_ptr<float>(), (int)cp.size(0), weight.data_ptr<float>(),
+            pp, NB, weight.data_ptr<float>(),
             bias.data_ptr<float>(), running_mean.data_ptr<float>(),
             running_var.data_ptr<float>(), save_mean.data_ptr<float>(),
             save_invstd.data_ptr<float>(), ws.data_ptr<float>(), rows,

@@ -419,6 +419,59 @@ void bn_bwd_stats_launch(const void* gout, const void* x, const void* out,
     }
 }
 
+// Pre-fold for conv-epilogue partials: mtiles can reach ~4700 for the
+// layer-1 shapes, and the single finalize launch only runs C/64 blocks —
+// one CU streaming ~2.4 MB serially (~15-20 us per BN measured as the
+// bn_*_reduce_finalize rows). Fold to NB2 rows first at full-grid width.
+__global__ __launch_bounds__(256) void bn_partial_prefold_kernel(
+    const float* __restrict__ part, float* __restrict__ out, int NB,
+    int NB2, int chunkC) {
+    __shared__ float sm[2 * 1024];
+    const int lane = threadIdx.x & 15;
+    const int sub = threadIdx.x >> 4;
+    const int cc = (blockIdx.x * 16 + lane) * 4;
+    const int g = blockIdx.y;
+    float4 sv = {0, 0, 0, 0}, qv = {0, 0, 0, 0};
+    for (int b = g + sub * NB2; b < NB; b += 16 * NB2) {
+        const float* p = part + (int64_t)b * 2 * chunkC;
+        const float4 a = *(const float4*)(p + cc);
+        const float4 z = *(const float4*)(p + chunkC + cc);
+        sv.x += a.x; sv.y += a.y; sv.z += a.z; sv.w += a.w;
+        qv.x += z.x; qv.y += z.y; qv.z += z.z; qv.w += z.w;
+    }
+    *(float4*)(sm + (sub * 16 + lane) * 4) = sv;
+    *(float4*)(sm + 1024 + (sub * 16 + lane) * 4) = qv;
+    __syncthreads();
+    #pragma unroll
+    for (int off = 8; off > 0; off >>= 1) {
+        if (sub < off) {
+            #pragma unroll
+            for (int k = 0; k < 4; ++k) {
+                sm[(sub * 16 + lane) * 4 + k] +=
+                    sm[((sub + off) * 16 + lane) * 4 + k];
+                sm[1024 + (sub * 16 + lane) * 4 + k] +=
+                    sm[1024 + ((sub + off) * 16 + lane) * 4 + k];
+            }
+        }
+        __syncthreads();
+    }
+    if (sub == 0) {
+        float* o = out + (int64_t)g * 2 * chunkC;
+        #pragma unroll
+        for (int k = 0; k < 4; ++k) {
+            o[cc + k] = sm[(lane * 4 + k)];
+            o[chunkC + cc + k] = sm[1024 + lane * 4 + k];
+        }
+    }
+}
+
+void bn_partial_prefold_launch(const float* part, float* out, int NB,
+                               int NB2, int chunkC, hipStream_t s) {
+    dim3 grid((unsigned)(chunkC / 64), (unsigned)NB2);
+    hipLaunchKernelGGL(bn_partial_prefold_kernel, grid, dim3(256), 0, s,
+                       part, out, NB, NB2, chunkC);
+}
+
 void bn_finalize_from_partials_launch(
     const float* part, int NB, const float* weight, const float* bias,
     float* rm, float* rv, float* save_mean, float* save_invstd, float* ws,

@@ -32,6 +32,9 @@ void bn_stats_launch(const void* x, float* ws, float* part,
                      float* running_mean, float* running_var, float* save_mean,
                      float* save_invstd, int64_t rows, int C, float momentum,
                      float eps, DT dt, hipStream_t s);
+// pre-fold big conv-epilogue partial sets to NB2 rows at full grid width
+void bn_partial_prefold_launch(const float* part, float* out, int NB,
+                               int NB2, int chunkC, hipStream_t s);
 // finalize from conv-epilogue partials ([NB][2][C])
 void bn_finalize_from_partials_launch(
     const float* part, int NB, const float* weight, const float* bias,
