@@ -75,7 +75,7 @@ constexpr int BK = 64;
 // configs (256 threads, 2 blocks/CU), 8 for the 256x128 big tile (512
 // threads, 1 block/CU — fewer glds per wave per K-step: 6 vs 8, same 32
 // MFMAs, so a higher MFMA:staging ratio on the deep-K layers).
-template <int MODE, int BM, int BN, int WN, int NBUF = 2>
+template <int MODE, int BM, int BN, int WN, int NBUF = 2, int BKT = BK>
 __global__ __launch_bounds__((BM / 64) * (BN / 64) * 64,
                              512 / ((BM / 64) * (BN / 64) * 64))
 void conv_igemm_kernel(
@@ -106,11 +106,21 @@ void conv_igemm_kernel(
            CUDAFunctor_add pass disappears (ops/conv.py junction stash) */
     ) {
     constexpr int NW = (BM / 64) * (BN / 64);   // waves per block
-    constexpr int A_ELEMS = BM * BK;
-    constexpr int B_ELEMS = BN * BK;
+    constexpr int A_ELEMS = BM * BKT;
+    constexpr int B_ELEMS = BN * BKT;
     constexpr int BUF_ELEMS = A_ELEMS + B_ELEMS;
-    constexpr int AI = BM / (8 * NW);    // A glds per wave per tile
-    constexpr int BI = BN / (8 * NW);    // B glds per wave per tile
+    // one glds = 64 lanes x 16 B = 512/BKT rows of BKT elements
+    constexpr int RPG = 512 / BKT;       // rows per glds (8 @64, 16 @32)
+    constexpr int GPR = BKT / 8;         // 16-B granules per row (8 / 4)
+    constexpr int KH = BKT / 32;         // mfma k-halves per staged tile
+    // granule swizzle: BK64 keeps the proven row&7; BK32 folds row>>2 in
+    // (plain row&3 repeats banks every 4 rows: 4-way b128 conflicts; the
+    // fold reaches the 2-way floor of 4 slots over 8 same-phase rows)
+    auto swz = [](int row) {
+        return BKT == 64 ? (row & 7) : ((row ^ (row >> 2)) & (GPR - 1));
+    };
+    constexpr int AI = BM / (RPG * NW);  // A glds per wave per tile
+    constexpr int BI = BN / (RPG * NW);  // B glds per wave per tile
 
     const int OC = (MODE == CONV_DGRAD) ? C : K;
     const int RC = (MODE == CONV_FWD) ? C : ((MODE == CONV_STEM) ? 64 : K);
@@ -150,10 +160,10 @@ void conv_igemm_kernel(
     long a_pix[AI];
     int a_hb[AI], a_wb[AI];
     bool a_mok[AI];
-    const int cslot = lane & 7;
+    const int cslot = lane % GPR;
     #pragma unroll
     for (int i = 0; i < AI; ++i) {
-        const int row = (wid * AI + i) * 8 + (lane >> 3);
+        const int row = (wid * AI + i) * RPG + lane / GPR;
         a_row[i] = row;
         const long m = m0 + row;
         const bool mok = m < M;
@@ -178,9 +188,10 @@ void conv_igemm_kernel(
     }
     int b_row[BI];
     #pragma unroll
-    for (int i = 0; i < BI; ++i) b_row[i] = (wid * BI + i) * 8 + (lane >> 3);
+    for (int i = 0; i < BI; ++i)
+        b_row[i] = (wid * BI + i) * RPG + lane / GPR;
 
-    const int cblocks = RC / BK;
+    const int cblocks = RC / BKT;
     const int T = nR * nS * cblocks;
 
     // Incremental staging offsets: stage() is called with strictly
@@ -198,10 +209,10 @@ void conv_igemm_kernel(
         unsigned short* base = lds + buf * BUF_ELEMS;
         if (rsi != last_rsi) {
             last_rsi = rsi;
-            const int cb = (it % cblocks) * BK;   // 0 except NBUF>1 prologue
+            const int cb = (it % cblocks) * BKT;  // 0 except NBUF>1 prologue
             #pragma unroll
             for (int i = 0; i < AI; ++i) {
-                const int cs = (cslot ^ (a_row[i] & 7)) * 8;
+                const int cs = (cslot ^ swz(a_row[i])) * 8;
                 bool ok = a_mok[i];
                 long off = 0;
                 if (MODE == CONV_STEM) {
@@ -224,7 +235,7 @@ void conv_igemm_kernel(
             #pragma unroll
             for (int i = 0; i < BI; ++i) {
                 const int row = b_row[i];
-                const int cs = (cslot ^ (row & 7)) * 8;
+                const int cs = (cslot ^ swz(row)) * 8;
                 if (MODE == CONV_STEM)
                     b_goff[i] = ((long)(n0 + row) * R + ri) * 64 + cs;
                 else if (MODE == CONV_FWD)
@@ -234,40 +245,41 @@ void conv_igemm_kernel(
             }
         } else {
             #pragma unroll
-            for (int i = 0; i < AI; ++i) a_goff[i] += BK;
+            for (int i = 0; i < AI; ++i) a_goff[i] += BKT;
             #pragma unroll
-            for (int i = 0; i < BI; ++i) b_goff[i] += BK;
+            for (int i = 0; i < BI; ++i) b_goff[i] += BKT;
         }
         #pragma unroll
         for (int i = 0; i < AI; ++i) {
             const unsigned short* sp = a_okc[i] ? src + a_goff[i] : conv_zero16;
-            FDA_GLDS16(sp, base + (wid * AI + i) * 8 * BK);
+            FDA_GLDS16(sp, base + (wid * AI + i) * RPG * BKT);
         }
         #pragma unroll
         for (int i = 0; i < BI; ++i) {
-            FDA_GLDS16(wgt + b_goff[i], base + A_ELEMS + (wid * BI + i) * 8 * BK);
+            FDA_GLDS16(wgt + b_goff[i],
+                       base + A_ELEMS + (wid * BI + i) * RPG * BKT);
         }
     };
 
     // ---- fragment read offsets (elements into an lds buffer) -------------
-    int a_off[4][2], b_off[4][2];
+    int a_off[4][KH], b_off[4][KH];
     {
         const int fr = lane & 15, fq = lane >> 4;
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             #pragma unroll
-            for (int kh = 0; kh < 2; ++kh) {
+            for (int kh = 0; kh < KH; ++kh) {
                 const int row = wm * 64 + mi * 16 + fr;
-                const int slot = (kh * 4 + fq) ^ (row & 7);
-                a_off[mi][kh] = row * BK + slot * 8;
+                const int slot = ((kh * 4 + fq) ^ swz(row)) & (GPR - 1);
+                a_off[mi][kh] = row * BKT + slot * 8;
             }
         #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
             #pragma unroll
-            for (int kh = 0; kh < 2; ++kh) {
+            for (int kh = 0; kh < KH; ++kh) {
                 const int row = wn * 64 + ni * 16 + fr;
-                const int slot = (kh * 4 + fq) ^ (row & 7);
-                b_off[ni][kh] = A_ELEMS + row * BK + slot * 8;
+                const int slot = ((kh * 4 + fq) ^ swz(row)) & (GPR - 1);
+                b_off[ni][kh] = A_ELEMS + row * BKT + slot * 8;
             }
     }
 
@@ -295,7 +307,9 @@ void conv_igemm_kernel(
         // reading before it arrived here. With NBUF==3 the counted wait
         // leaves tile it+1's DMA in flight across the barrier (T4).
         if (NBUF > 2 && it + 1 < itN) {
-            if constexpr (GPW == 6)
+            if constexpr (GPW == 4)
+                asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+            else if constexpr (GPW == 6)
                 asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
             else if constexpr (GPW == 8)
                 asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
@@ -314,9 +328,9 @@ void conv_igemm_kernel(
         // is required — see conv_wgrad_kernel.)
         if (it + NBUF - 1 < itN) stage((it + NBUF - 1) % NBUF, it + NBUF - 1);
         const unsigned short* buf = lds + (it % NBUF) * BUF_ELEMS;
-        short8 af[4][2], bf[4][2];
+        short8 af[4][KH], bf[4][KH];
         #pragma unroll
-        for (int kh = 0; kh < 2; ++kh) {
+        for (int kh = 0; kh < KH; ++kh) {
             #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
                 af[mi][kh] = *(const short8*)(buf + a_off[mi][kh]);
@@ -326,7 +340,7 @@ void conv_igemm_kernel(
         }
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
-        for (int kh = 0; kh < 2; ++kh)
+        for (int kh = 0; kh < KH; ++kh)
             #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
                 #pragma unroll
@@ -488,7 +502,7 @@ void conv_igemm_kernel(
     }
 }
 
-template <int MODE, int BM, int BN, int WN, int NBUF = 2>
+template <int MODE, int BM, int BN, int WN, int NBUF = 2, int BKT = BK>
 static void launch_cfg(const void* src, const void* wgt, void* out,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
@@ -504,17 +518,17 @@ static void launch_cfg(const void* src, const void* wgt, void* out,
     constexpr unsigned NTHREADS = (BM / 64) * (BN / 64) * 64;
     dim3 grid((unsigned)((M + BM - 1) / BM), (unsigned)(OC / BN),
               zbase * (unsigned)SK);
-    const size_t shmem = NBUF * (BM * BK + BN * BK) * sizeof(unsigned short);
+    const size_t shmem = NBUF * (BM * BKT + BN * BKT) * sizeof(unsigned short);
     if (shmem > 65536) {
         static bool raised = [] {
             hipFuncSetAttribute(
-                (const void*)&conv_igemm_kernel<MODE, BM, BN, WN, NBUF>,
+                (const void*)&conv_igemm_kernel<MODE, BM, BN, WN, NBUF, BKT>,
                 hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
             return true;
         }();
         (void)raised;
     }
-    hipLaunchKernelGGL((conv_igemm_kernel<MODE, BM, BN, WN, NBUF>), grid,
+    hipLaunchKernelGGL((conv_igemm_kernel<MODE, BM, BN, WN, NBUF, BKT>), grid,
                        dim3(NTHREADS), shmem, stream,
                        (const unsigned short*)src,
                        (const unsigned short*)wgt, (unsigned short*)out,
@@ -627,8 +641,12 @@ __global__ __launch_bounds__(512, 1) void conv_igemm8_kernel(
     float* __restrict__ stats,
     const unsigned short* __restrict__ accsrc) {
     constexpr int BM = 256, BN = 128, NBUF = 3;
-    constexpr int A_ELEMS = BM * BK;
-    constexpr int B_ELEMS = BN * BK;
+    constexpr int BKT = BK;              // this variant is BK=64 only
+    constexpr int RPG = 8, GPR = 8;      // rows / 16-B granules per glds
+    constexpr int KH = 2;                // mfma k-halves (BK=64)
+    auto swz = [](int row) { return row & 7; };
+    constexpr int A_ELEMS = BM * BKT;
+    constexpr int B_ELEMS = BN * BKT;
     constexpr int BUF_ELEMS = A_ELEMS + B_ELEMS;
     constexpr int AI = 4, BI = 2;        // glds per wave per tile
     constexpr int NW = 8, WN = 2;
@@ -667,10 +685,10 @@ __global__ __launch_bounds__(512, 1) void conv_igemm8_kernel(
     long a_pix[AI];
     int a_hb[AI], a_wb[AI];
     bool a_mok[AI];
-    const int cslot = lane & 7;
+    const int cslot = lane % GPR;
     #pragma unroll
     for (int i = 0; i < AI; ++i) {
-        const int row = (wid * AI + i) * 8 + (lane >> 3);
+        const int row = (wid * AI + i) * RPG + lane / GPR;
         a_row[i] = row;
         const long m = m0 + row;
         const bool mok = m < M;
@@ -691,9 +709,10 @@ __global__ __launch_bounds__(512, 1) void conv_igemm8_kernel(
     }
     int b_row[BI];
     #pragma unroll
-    for (int i = 0; i < BI; ++i) b_row[i] = (wid * BI + i) * 8 + (lane >> 3);
+    for (int i = 0; i < BI; ++i)
+        b_row[i] = (wid * BI + i) * RPG + lane / GPR;
 
-    const int cblocks = RC / BK;
+    const int cblocks = RC / BKT;
     const int T = nR * nS * cblocks;
 
     long a_goff[AI];
@@ -728,7 +747,7 @@ __global__ __launch_bounds__(512, 1) void conv_igemm8_kernel(
             #pragma unroll
             for (int i = 0; i < BI; ++i) {
                 const int row = b_row[i];
-                const int cs = (cslot ^ (row & 7)) * 8;
+                const int cs = (cslot ^ (row & (GPR - 1))) * 8;
                 if (MODE == CONV_FWD)
                     b_goff[i] = (long)(n0 + row) * R * S * C + (long)rs * C + cb + cs;
                 else
@@ -743,24 +762,24 @@ __global__ __launch_bounds__(512, 1) void conv_igemm8_kernel(
     };
 
     // fragment read offsets
-    int a_off[4][2], b_off[4][2];
+    int a_off[4][KH], b_off[4][KH];
     {
         const int fr = lane & 15, fq = lane >> 4;
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             #pragma unroll
-            for (int kh = 0; kh < 2; ++kh) {
+            for (int kh = 0; kh < KH; ++kh) {
                 const int row = wm * 64 + mi * 16 + fr;
-                const int slot = (kh * 4 + fq) ^ (row & 7);
-                a_off[mi][kh] = row * BK + slot * 8;
+                const int slot = ((kh * 4 + fq) ^ swz(row)) & (GPR - 1);
+                a_off[mi][kh] = row * BKT + slot * 8;
             }
         #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
             #pragma unroll
-            for (int kh = 0; kh < 2; ++kh) {
+            for (int kh = 0; kh < KH; ++kh) {
                 const int row = wn * 64 + ni * 16 + fr;
-                const int slot = (kh * 4 + fq) ^ (row & 7);
-                b_off[ni][kh] = A_ELEMS + row * BK + slot * 8;
+                const int slot = ((kh * 4 + fq) ^ swz(row)) & (GPR - 1);
+                b_off[ni][kh] = A_ELEMS + row * BKT + slot * 8;
             }
     }
 
@@ -977,6 +996,17 @@ static int conv_nbuf() {
     return v;
 }
 
+static int conv_bk32() {
+    // BK=32 x NBUF=3 for the 128x128 config: 48 KB LDS/block -> 3
+    // blocks/CU (12 waves) vs the default's 2 — the one untried point in
+    // the occupancy/pipeline-depth space (A/B knob FLUXDIST_CONV_BK32).
+    static int v = [] {
+        const char* e = getenv("FLUXDIST_CONV_BK32");
+        return e ? atoi(e) : 0;
+    }();
+    return v;
+}
+
 static int conv_bigtile() {
     // 256x128 512-thread tile (8 waves, 6 glds/wave/K-step at the same 32
     // MFMAs). A/B knob FLUXDIST_CONV_BIGTILE: 0 = never, 1 = whenever
@@ -1081,7 +1111,11 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
         return;
     }
     if (dgrad) {
-        if (big && nb3)
+        if (big && conv_bk32())
+            launch_cfg<CONV_DGRAD, 128, 128, 2, 3, 32>(
+                src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
+                stream, nullptr, skpart, SK, cnt, accsrc);
+        else if (big && nb3)
             launch_cfg<CONV_DGRAD, 128, 128, 2, 3>(src, wgt, out, N, H, W, C,
                                                    K, P, Q, R, S, sy, sx, py,
                                                    px, stream, nullptr,
@@ -1097,7 +1131,11 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                                                stream, nullptr, skpart, SK,
                                                cnt, accsrc);
     } else {
-        if (big && nb3)
+        if (big && conv_bk32())
+            launch_cfg<CONV_FWD, 128, 128, 2, 3, 32>(
+                src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
+                stream, stats, skpart, SK, cnt, accsrc);
+        else if (big && nb3)
             launch_cfg<CONV_FWD, 128, 128, 2, 3>(src, wgt, out, N, H, W, C,
                                                  K, P, Q, R, S, sy, sx, py,
                                                  px, stream, stats, skpart,
