@@ -998,11 +998,14 @@ static int conv_nbuf() {
 
 static int conv_bk32() {
     // BK=32 x NBUF=3 for the 128x128 config: 48 KB LDS/block -> 3
-    // blocks/CU (12 waves) vs the default's 2 — the one untried point in
-    // the occupancy/pipeline-depth space (A/B knob FLUXDIST_CONV_BK32).
+    // blocks/CU (12 waves) vs the default's 2. Measured (r2 A/B,
+    // profiles/ab_conv8.md addendum): -9..-28% on the big-M layer-1/2
+    // legs, +17% on the small-M 14x14 legs (294 blocks can't feed 3
+    // blocks/CU), so dispatch is grid-size-gated below. 0 = never,
+    // 1 = gated by fill (default), 2 = force everywhere (A/B).
     static int v = [] {
         const char* e = getenv("FLUXDIST_CONV_BK32");
-        return e ? atoi(e) : 0;
+        return e ? atoi(e) : 1;
     }();
     return v;
 }
@@ -1110,8 +1113,14 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                 stream, stats, skpart, SK, cnt, accsrc);
         return;
     }
+    // the 3-blocks/CU BK32 config needs >=~576 blocks to fill; below
+    // that the lost per-block depth outweighs the extra block overlap
+    const long blocks128 =
+        ((Mv + 127) / 128) * (OC / 128) * (long)zbase * SK;
+    const bool bk32 = big && conv_bk32() &&
+                      (conv_bk32() >= 2 || blocks128 >= 576);
     if (dgrad) {
-        if (big && conv_bk32())
+        if (bk32)
             launch_cfg<CONV_DGRAD, 128, 128, 2, 3, 32>(
                 src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
                 stream, nullptr, skpart, SK, cnt, accsrc);
@@ -1131,7 +1140,7 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                                                stream, nullptr, skpart, SK,
                                                cnt, accsrc);
     } else {
-        if (big && conv_bk32())
+        if (bk32)
             launch_cfg<CONV_FWD, 128, 128, 2, 3, 32>(
                 src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
                 stream, stats, skpart, SK, cnt, accsrc);
