@@ -996,7 +996,7 @@ static int conv_nbuf() {
     return v;
 }
 
-static int conv_bk32() {
+int conv_bk32_knob() {
     // BK=32 x NBUF=3 for the 128x128 config: 48 KB LDS/block -> 3
     // blocks/CU (12 waves) vs the default's 2. Measured (r2 A/B,
     // profiles/ab_conv8.md addendum): -9..-28% on the big-M layer-1/2
@@ -1008,6 +1008,15 @@ static int conv_bk32() {
         return e ? atoi(e) : 1;
     }();
     return v;
+}
+
+int conv_bk32_check(long blocks64);
+
+int conv_bk32_check(long blocks64) {
+    // shared gate for plan + launcher (the 128x64 BK32 config)
+    extern int conv_bk32_knob();
+    const int v = conv_bk32_knob();
+    return v && (v >= 2 || blocks64 >= 1024);
 }
 
 static int conv_bigtile() {
@@ -1070,6 +1079,13 @@ void conv_igemm_plan(long M, int OC, long T, int zbase,
         const long blocks256 = ((M + 255) / 256) * (OC / 128) * zbase;
         if (blocks256 >= 256) { BM = 256; BN = 128; }
     }
+    if (!big) {
+        // 64-channel shapes may take the 128x64 x BK32 config (4 blocks/
+        // CU); BM=128 here so the stats workspace matches the launch
+        const long blocks64 = ((M + 127) / 128) * (OC / 64) * zbase * SK;
+        extern int conv_bk32_check(long blocks64);
+        if (conv_bk32_check(blocks64)) BM = 128;
+    }
     *bm = BM; *bn = BN; *sk = SK;
 }
 
@@ -1117,8 +1133,12 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
     // that the lost per-block depth outweighs the extra block overlap
     const long blocks128 =
         ((Mv + 127) / 128) * (OC / 128) * (long)zbase * SK;
-    const bool bk32 = big && conv_bk32() &&
-                      (conv_bk32() >= 2 || blocks128 >= 576);
+    const bool bk32 = big && conv_bk32_knob() &&
+                      (conv_bk32_knob() >= 2 || blocks128 >= 576);
+    // 64-channel shapes (layer 1): 128x64 x BK32 x NBUF3 = 36 KB LDS ->
+    // 4 blocks/CU of 2 waves; the plan already switched BM to 128 when
+    // the grid qualifies (stats workspace sizing must match)
+    const bool bk32s = !big && BM == 128;
     if (dgrad) {
         if (bk32)
             launch_cfg<CONV_DGRAD, 128, 128, 2, 3, 32>(
@@ -1134,6 +1154,10 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                                                 P, Q, R, S, sy, sx, py, px,
                                                 stream, nullptr, skpart, SK,
                                                 cnt, accsrc);
+        else if (bk32s)
+            launch_cfg<CONV_DGRAD, 128, 64, 1, 3, 32>(
+                src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
+                stream, nullptr, skpart, SK, cnt, accsrc);
         else
             launch_cfg<CONV_DGRAD, 256, 64, 1>(src, wgt, out, N, H, W, C, K,
                                                P, Q, R, S, sy, sx, py, px,
@@ -1154,6 +1178,10 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                                               P, Q, R, S, sy, sx, py, px,
                                               stream, stats, skpart, SK,
                                               cnt);
+        else if (bk32s)
+            launch_cfg<CONV_FWD, 128, 64, 1, 3, 32>(
+                src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
+                stream, stats, skpart, SK, cnt, accsrc);
         else
             launch_cfg<CONV_FWD, 256, 64, 1>(src, wgt, out, N, H, W, C, K,
                                              P, Q, R, S, sy, sx, py, px,

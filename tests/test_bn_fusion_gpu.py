@@ -19,6 +19,7 @@ if not torch.cuda.is_available():  # pragma: no cover
     (4, 64, 56, 56, 64, 3, 1),
     (96, 256, 14, 14, 512, 3, 2),   # split-K fwd shape (combine emits stats)
     (4, 3, 64, 64, 64, 7, 2),       # stem kernel stats
+    (96, 64, 56, 56, 64, 3, 1),     # full-batch layer1: BK32 128x64 stats
 ])
 def test_conv_bn_fused_stats_match_unfused(shape, monkeypatch):
     from fluxdistributed_amd.models.resnet import FusedBNAct

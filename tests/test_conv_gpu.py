@@ -37,6 +37,8 @@ SHAPES = [
                                    # (M=4704 -> wgrad takes the FT=4 tile)
     (2, 1024, 14, 14, 2048, 1, 2), # layer4 downsample 1x1 s2 K=2048
     (4, 512, 14, 14, 512, 3, 2),   # r50 layer4 3x3 s2 C=K=512
+    (96, 64, 56, 56, 64, 3, 1),    # full-batch layer1 (BK32 128x64 config)
+    (96, 128, 28, 28, 128, 3, 1),  # full-batch layer2 (BK32 128x128 config)
 ]
 
 
