@@ -1013,10 +1013,14 @@ int conv_bk32_knob() {
 int conv_bk32_check(long blocks64);
 
 int conv_bk32_check(long blocks64) {
-    // shared gate for plan + launcher (the 128x64 BK32 config)
+    // 128x64 x BK32 gate — measured a large LOSS on layer 1 (fwd 61->99,
+    // dgrad 76->100 us: the 2-wave block drops the MFMA:glds ratio to
+    // 2.7:1 and these legs are staging-bandwidth-bound). Only the forced
+    // A/B mode (FLUXDIST_CONV_BK32=2) takes it; the 128x128 BK32 config
+    // keeps its own fill gate (the measured +3.8% win).
     extern int conv_bk32_knob();
-    const int v = conv_bk32_knob();
-    return v && (v >= 2 || blocks64 >= 1024);
+    (void)blocks64;
+    return conv_bk32_knob() >= 2;
 }
 
 static int conv_bigtile() {
