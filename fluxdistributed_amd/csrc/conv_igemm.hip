@@ -1613,11 +1613,17 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
                        int R, int S, int sy, int sx, int py, int px,
                        hipStream_t stream) {
     const long M = (long)N * P * Q;
+    // A/B: FLUXDIST_WGRAD_FT2=1 forces the 64x64 tile everywhere (its
+    // 48 KB 3-ring fits 3 blocks/CU vs FT4's 2 — the BK32 lesson)
+    static const bool force_ft2 = [] {
+        const char* e = getenv("FLUXDIST_WGRAD_FT2");
+        return e && e[0] == '1';
+    }();
     // FT=4 quarters the block count; only worth it when the grid still
     // fills the 256 CUs (small-M 1x1 shapes measured 1.4x slower on it)
     const long tiles4 = (long)(K / 128) * (C / 128) * R * S;
     const long blocks4_max = tiles4 * ((M + WG_MCH - 1) / WG_MCH);
-    if (K % 128 == 0 && C % 128 == 0 && blocks4_max >= 192) {
+    if (!force_ft2 && K % 128 == 0 && C % 128 == 0 && blocks4_max >= 192) {
         const int mch = pick_mch(M, tiles4, 768);
         const int nch = (int)((M + mch - 1) / mch);
         dim3 grid((unsigned)(K / 128), (unsigned)(C / 128),
