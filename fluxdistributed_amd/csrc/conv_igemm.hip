@@ -1274,7 +1274,7 @@ typedef __attribute__((ext_vector_type(4))) short short4_;
 constexpr int WG_BM = 64;      // m per K-step
 constexpr int WG_MCH = 2048;   // pixels per block (chunk)
 
-template <int FT, bool STEM = false, bool ASMRD = false>
+template <int FT, bool STEM = false, bool ASMRD = false, int MB = WG_BM>
 __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
     const unsigned short* __restrict__ dy,   // [M][K] (NHWC out grad)
     const unsigned short* __restrict__ x,    // [N,H,W,C]
@@ -1282,7 +1282,10 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
     int N, int H, int W, int C, int K, int P, int Q,
     int R, int S, int sy, int sx, int py, int px, int nch, int mch) {
     constexpr int TCH = 32 * FT;             // tile channels per operand
-    constexpr int TILE_ELEMS = WG_BM * TCH;  // one operand tile
+    constexpr int TILE_ELEMS = MB * TCH;     // one operand tile
+    constexpr int GI = MB * TCH / 2048;      // glds per wave per operand
+    constexpr int NWIN = MB / 4;             // tr16 windows (4 m x 16 ch)
+    constexpr int KSN = MB / 32;             // MFMA m-halves per K-step
     const int rs = blockIdx.z / nch;
     const int chunk = blockIdx.z % nch;
     const int r = rs / S, s = rs % S;
@@ -1311,21 +1314,23 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
     // called with strictly sequential m-bases (prologue 0,1 then it+2), so
     // each lane tracks its pixel (n,p,q) with constant-delta carries — no
     // per-iteration division (those were ~3x the MFMA issue time here).
-    int st_q[FT], st_p[FT], st_n[FT], st_ch[FT];
-    long st_dyoff[FT];
-    long st_m[FT];
+    int st_q[GI], st_p[GI], st_n[GI], st_ch[GI];
+    long st_dyoff[GI];
+    long st_m[GI];
     #pragma unroll
-    for (int i = 0; i < FT; ++i) {
-        const int ln = (wid * FT + i) * 64 + lane;
+    for (int i = 0; i < GI; ++i) {
+        const int ln = (wid * GI + i) * 64 + lane;
         const int e8 = ln * 8;
-        const int kb = e8 >> 10;
-        const int rr = e8 & 1023;
-        // window position -> window index: evens at pos 0-7, odds at 8-15,
-        // so the two windows a half-wave tr-reads simultaneously sit on
-        // different 128-B bank halves (2-way conflict measured at 6.3% of
-        // wave cycles with the linear layout)
+        const int kb = e8 / (MB * 16);
+        const int rr = e8 % (MB * 16);
+        // window position -> window index: evens in the low half of the
+        // positions, odds in the high half, so the two windows a
+        // half-wave tr-reads simultaneously sit on different 128-B bank
+        // halves (2-way conflict measured at 6.3% of wave cycles with
+        // the linear layout)
         const int pos = rr >> 6;
-        const int v = (pos < 8) ? pos * 2 : (pos - 8) * 2 + 1;
+        const int v = (pos < NWIN / 2) ? pos * 2
+                                       : (pos - NWIN / 2) * 2 + 1;
         const int ml = (v << 2) + ((rr & 63) >> 4);
         const int ch = (kb << 4) + (rr & 15);
         st_ch[i] = ch;
@@ -1336,19 +1341,19 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
         st_n[i] = (int)(m / ((long)Q * P));
         st_dyoff[i] = m * K + k0 + ch;
     }
-    const int dQ = WG_BM % Q, dP = (WG_BM / Q) % P, dN0 = WG_BM / (Q * P);
-    const long dDY = (long)WG_BM * K;
+    const int dQ = MB % Q, dP = (MB / Q) % P, dN0 = MB / (Q * P);
+    const long dDY = (long)MB * K;
 
     auto stage = [&](int buf) {
         unsigned short* base = lds + buf * 2 * TILE_ELEMS;
         #pragma unroll
-        for (int i = 0; i < FT; ++i) {
+        for (int i = 0; i < GI; ++i) {
             const unsigned short* sp =
                 (st_m[i] < mend) ? dy + st_dyoff[i] : conv_zero16;
-            FDA_GLDS16(sp, base + (wid * FT + i) * 512);
+            FDA_GLDS16(sp, base + (wid * GI + i) * 512);
         }
         #pragma unroll
-        for (int i = 0; i < FT; ++i) {
+        for (int i = 0; i < GI; ++i) {
             const unsigned short* sp = conv_zero16;
             if (st_m[i] < mend) {
                 if (STEM) {
@@ -1365,13 +1370,13 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
                              c0 + st_ch[i];
                 }
             }
-            FDA_GLDS16(sp, base + TILE_ELEMS + (wid * FT + i) * 512);
+            FDA_GLDS16(sp, base + TILE_ELEMS + (wid * GI + i) * 512);
         }
-        // advance 64 pixels (bounded carries; dP < P, so p needs at most
+        // advance MB pixels (bounded carries; dP < P, so p needs at most
         // two conditional wraps after the q carry)
         #pragma unroll
-        for (int i = 0; i < FT; ++i) {
-            st_m[i] += WG_BM;
+        for (int i = 0; i < GI; ++i) {
+            st_m[i] += MB;
             st_dyoff[i] += dDY;
             int q = st_q[i] + dQ;
             int p = st_p[i] + dP;
@@ -1389,14 +1394,16 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
     // flight across the barrier while tile t computes (2*FT glds per wave
     // per tile-pair). FT=4 uses a 2-deep ring (96 KiB would exceed 1
     // block/CU headroom at 3).
-    constexpr int RING = (FT == 2) ? 3 : 2;
-    const int nsteps = (int)((mend - mb0 + WG_BM - 1) / WG_BM);
+    constexpr int RING = (FT == 2 || MB == 32) ? 3 : 2;
+    const int nsteps = (int)((mend - mb0 + MB - 1) / MB);
     if (nsteps > 0) stage(0);
     if (RING > 2 && nsteps > 1) stage(1);
     for (int it = 0; it < nsteps; ++it) {
         if (RING > 2 && it + 1 < nsteps) {
-            if (FT == 2) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-            else         asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+            if constexpr (2 * GI == 4)
+                asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
         } else {
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
@@ -1411,7 +1418,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
         // drain wait only for data this iteration needs anyway, and the
         // DMA now flies over the MFMA cluster.
         const unsigned short* buf = lds + (it % RING) * 2 * TILE_ELEMS;
-        short4_ a[FT][2][2], b[FT][2][2];   // [fi][ks][half]
+        short4_ a[FT][KSN][2], b[FT][KSN][2];   // [fi][ks][half]
         if constexpr (ASMRD) {
             // tr16 reads via inline asm, INVISIBLE to hipcc's wait
             // inserter: the compiler otherwise (a) force-drains all
@@ -1425,17 +1432,18 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
             if (it + RING - 1 < nsteps) stage((it + RING - 1) % RING);
             __builtin_amdgcn_sched_barrier(0);
             #pragma unroll
-            for (int ks = 0; ks < 2; ++ks)
+            for (int ks = 0; ks < KSN; ++ks)
                 #pragma unroll
                 for (int fi = 0; fi < FT; ++fi) {
                     const int kb_a = wk * FT + fi;
                     const int kb_b = wc * FT + fi;
                     const int v0 = ks * 8 + lg * 2;
                     const int p0 = (v0 >> 1);
-                    const int p1 = p0 + 8;
-                    const unsigned short* pa = buf + kb_a * 1024 + l15 * 4;
+                    const int p1 = p0 + NWIN / 2;
+                    const unsigned short* pa =
+                        buf + kb_a * (MB * 16) + l15 * 4;
                     const unsigned short* pb =
-                        buf + TILE_ELEMS + kb_b * 1024 + l15 * 4;
+                        buf + TILE_ELEMS + kb_b * (MB * 16) + l15 * 4;
                     const unsigned aa0 = (unsigned)(unsigned long)
                         (const __attribute__((address_space(3))) char*)
                         (const char*)(pa + p0 * 64);
@@ -1459,14 +1467,14 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
                 }
             __builtin_amdgcn_s_setprio(1);
             #pragma unroll
-            for (int ks = 0; ks < 2; ++ks) {
+            for (int ks = 0; ks < KSN; ++ks) {
                 // sched_barrier(0) pins the MFMA clusters BETWEEN the
                 // counted waits: MFMAs are pure-register ops, so without
                 // it LLVM floats both clusters past both waits — and the
                 // asm-read results carry no wait-dependency at all (the
                 // hand-counted waits ARE the only correctness fence).
                 __builtin_amdgcn_sched_barrier(0);
-                if (ks == 0) {
+                if (ks == 0 && KSN == 2) {
                     // lgkmcnt is 4-bit (max 15): for FT=4 the ideal
                     // "leave 16 in flight" clamps to 15 (ds_reads return
                     // in order, so this waits one extra ks=1 read)
@@ -1499,15 +1507,16 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
         #pragma unroll
         for (int fi = 0; fi < FT; ++fi)
             #pragma unroll
-            for (int ks = 0; ks < 2; ++ks) {
+            for (int ks = 0; ks < KSN; ++ks) {
                 const int kb_a = wk * FT + fi;
                 const int kb_b = wc * FT + fi;
                 const int v0 = ks * 8 + lg * 2;          // window indices
-                const int p0 = (v0 >> 1);                // v0 even -> pos 0-7
-                const int p1 = p0 + 8;                   // v0+1 odd -> pos 8-15
-                const unsigned short* pa = buf + kb_a * 1024 + l15 * 4;
+                const int p0 = (v0 >> 1);                // v0 even -> low half
+                const int p1 = p0 + NWIN / 2;            // v0+1 odd -> high
+                const unsigned short* pa =
+                    buf + kb_a * (MB * 16) + l15 * 4;
                 const unsigned short* pb =
-                    buf + TILE_ELEMS + kb_b * 1024 + l15 * 4;
+                    buf + TILE_ELEMS + kb_b * (MB * 16) + l15 * 4;
                 a[fi][ks][0] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
                     (__attribute__((address_space(3))) short4_*)(pa + p0 * 64));
                 a[fi][ks][1] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
@@ -1520,7 +1529,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
         if (it + RING - 1 < nsteps) stage((it + RING - 1) % RING);
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
-        for (int ks = 0; ks < 2; ++ks)
+        for (int ks = 0; ks < KSN; ++ks)
             #pragma unroll
             for (int ki = 0; ki < FT; ++ki)
                 #pragma unroll
@@ -1628,6 +1637,21 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
         const int nch = (int)((M + mch - 1) / mch);
         dim3 grid((unsigned)(K / 128), (unsigned)(C / 128),
                   (unsigned)(R * S * nch));
+        static const bool mb32 = [] {
+            // A/B knob: FT4 with a 32-pixel K-step — 48 KB 3-ring at 3
+            // blocks/CU, same 4:1 MFMA:glds ratio (the BK32 recipe)
+            const char* e = getenv("FLUXDIST_WGRAD_MB32");
+            return e && e[0] == '1';
+        }();
+        if (mb32) {
+            const size_t shmem = 3 * 2 * (32 * 128) * sizeof(unsigned short);
+            hipLaunchKernelGGL((conv_wgrad_kernel<4, false, false, 32>),
+                               grid, dim3(256), shmem, stream,
+                               (const unsigned short*)dy,
+                               (const unsigned short*)x, ws, N, H, W, C, K,
+                               P, Q, R, S, sy, sx, py, px, nch, mch);
+            return;
+        }
         const size_t shmem = 2 * 2 * (WG_BM * 128) * sizeof(unsigned short);
         if (wgrad_asm())
             hipLaunchKernelGGL((conv_wgrad_kernel<4, false, true>), grid,
