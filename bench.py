@@ -172,6 +172,8 @@ def main():
             "dtype": args.dtype,
             "data": "synthetic",
             "final_loss": round(float(loss.detach()), 4),
+            "peak_mem_gb": (round(torch.cuda.max_memory_allocated() / 2**30, 2)
+                            if device.type == "cuda" else None),
             "config": {
                 "model": args.model,
                 "global_batch": args.batch * max(world, 1),
