@@ -826,7 +826,7 @@ __global__ __launch_bounds__(512, 1) void conv_igemm8_kernel(
         // ---- phase 0: kh=0 (16 MFMAs) ----------------------------------
         // 2 phases of 16 MFMAs: the 4x8-MFMA grain measured ~10% SLOWER
         // (per-phase barrier cost not amortized at this tile size;
-        // profiles/ab_conv8.md) — FLUXDIST_CONV8_P4=1 selects it back.
+        // profiles/ab_conv8.md).
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             af0[mi] = *(const short8*)(rbuf + a_off[mi][0]);
