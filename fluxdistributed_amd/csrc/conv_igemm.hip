@@ -309,6 +309,8 @@ void conv_igemm_kernel(
         if (NBUF > 2 && it + 1 < itN) {
             if constexpr (GPW == 4)
                 asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+            else if constexpr (GPW == 5)
+                asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
             else if constexpr (GPW == 6)
                 asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
             else if constexpr (GPW == 8)
@@ -1143,6 +1145,13 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
     // 4 blocks/CU of 2 waves; the plan already switched BM to 128 when
     // the grid qualifies (stats workspace sizing must match)
     const bool bk32s = !big && BM == 128;
+    // 256x64 x BK32 x 3-ring: SAME 2 blocks/CU and the same 3.2:1
+    // MFMA:glds ratio as the BK64 default, but with a second tile in
+    // flight (60 KB x 2 fits; 3 x 80 KB at BK64 never did)
+    const long blocks256x64 =
+        ((Mv + 255) / 256) * (OC / 64) * (long)zbase * SK;
+    const bool bk32r = !big && BM == 256 && conv_bk32_knob() &&
+                       (conv_bk32_knob() >= 2 || blocks256x64 >= 512);
     if (dgrad) {
         if (bk32)
             launch_cfg<CONV_DGRAD, 128, 128, 2, 3, 32>(
@@ -1160,6 +1169,10 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                                                 cnt, accsrc);
         else if (bk32s)
             launch_cfg<CONV_DGRAD, 128, 64, 1, 3, 32>(
+                src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
+                stream, nullptr, skpart, SK, cnt, accsrc);
+        else if (bk32r)
+            launch_cfg<CONV_DGRAD, 256, 64, 1, 3, 32>(
                 src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
                 stream, nullptr, skpart, SK, cnt, accsrc);
         else
@@ -1184,6 +1197,10 @@ void conv_igemm_launch(const void* src, const void* wgt, void* out,
                                               cnt);
         else if (bk32s)
             launch_cfg<CONV_FWD, 128, 64, 1, 3, 32>(
+                src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
+                stream, stats, skpart, SK, cnt, accsrc);
+        else if (bk32r)
+            launch_cfg<CONV_FWD, 256, 64, 1, 3, 32>(
                 src, wgt, out, N, H, W, C, K, P, Q, R, S, sy, sx, py, px,
                 stream, stats, skpart, SK, cnt, accsrc);
         else
