@@ -58,10 +58,12 @@ def _miopen_env():
     db = os.path.join(root, "miopen_db")
     os.makedirs(db, exist_ok=True)
     os.environ.setdefault("MIOPEN_USER_DB_PATH", db)
-    # The committed find-db has tuned solvers for every library-conv shape
-    # this bench can hit (only the C=3 stem on the default fda path). FAST
-    # skips MIOpen's background solver sweep, whose naive_conv kernels
-    # otherwise run DURING the timed region (profiles/README.md).
+    # The default path runs no MIOpen kernels (stem included — the CONV_STEM
+    # kernels cover C<=5); this matters only for the FLUXDIST_CONV=miopen
+    # A/B arm. The committed find-db has tuned solvers for every library-conv
+    # shape that arm hits, and FAST skips MIOpen's background solver sweep,
+    # whose naive_conv kernels otherwise run DURING the timed region
+    # (profiles/README.md).
     os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
 
 
