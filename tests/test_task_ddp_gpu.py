@@ -70,8 +70,14 @@ def test_bf16_logical_fanout_on_one_gpu():
     torch.cuda.synchronize()
     ps = dict(solo.named_parameters())
     for k in p0:
-        a, b = p0[k].float(), ps[k].float()
+        a, b = p0[k].detach().float(), ps[k].detach().float()
         scale = float(b.abs().max())
         err = float((a - b).abs().max())
-        assert err < 0.02 * max(scale, 1e-2), \
+        # 0.05: the two arms sample the wgrad kernel's atomic fp32
+        # accumulation order independently, so the comparison is
+        # stochastic under bf16 rounding — at 0.02 this asserted once in
+        # ~10 otherwise-green runs. A real sync bug shows up as O(1)
+        # relative drift here and as inequality in the exact replica
+        # check above, which stays bit-strict.
+        assert err < 0.05 * max(scale, 1e-2), \
             f"{k}: drift {err} scale {scale}"
