@@ -28,6 +28,28 @@ def test_train_cli_task_mode(tmp_path):
     assert "model" in ck and "optimizer" in ck
 
 
+def test_train_cli_process_mode_world2(tmp_path):
+    """train.py process mode launched the way the driver launches it:
+    torchrun, 2 ranks, 127.0.0.1 rendezvous — gloo on CPU (the same
+    run_process() code path RCCL takes on GPUs)."""
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    ckdir = tmp_path / "w"
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--standalone",
+           "--local-addr", "127.0.0.1",
+           os.path.join(root, "train.py"),
+           "--mode", "process", "--model", "resnet18", "--small-input",
+           "--steps", "2", "--batch", "4", "--image-size", "32",
+           "--num-classes", "8", "--dtype", "fp32", "--data", "synthetic",
+           "--checkpoint-dir", str(ckdir), "--log-every", "1",
+           "--val-every", "0"]
+    r = subprocess.run(cmd, cwd=root, capture_output=True, text=True,
+                       timeout=600)
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    assert (ckdir / "resnet18_final.pt").exists(), \
+        list(ckdir.iterdir()) if ckdir.exists() else "no dir"
+
+
 def test_train_cli_resume(tmp_path):
     """--checkpoint-every writes a per-cycle checkpoint; --resume loads
     model AND optimizer state into every replica (train.py wiring, not
