@@ -457,6 +457,22 @@ at::Tensor conv_igemm_wgrad(at::Tensor dy, at::Tensor x,
     const int K = (int)dy.size(1), P = (int)dy.size(2), Q = (int)dy.size(3);
     TORCH_CHECK(C % 64 == 0 && K % 64 == 0);
     auto ws = at::zeros({K, R * S * C}, x.options().dtype(at::kFloat));
+    if (fda::conv_wgrad_two_phase()) {
+        int ft, mch, nch;
+        fda::conv_wgrad_plan((long)N * P * Q, C, K, (int)R, (int)S,
+                             &ft, &mch, &nch);
+        auto part = at::empty({(int64_t)nch, (int64_t)K * R * S * C},
+                              x.options().dtype(at::kFloat));
+        fda::conv_wgrad_launch(dy.data_ptr(), x.data_ptr(),
+                               ws.data_ptr<float>(), N, H, W, C, K, P, Q,
+                               (int)R, (int)S, (int)sy, (int)sx, (int)py,
+                               (int)px, cur_stream(),
+                               part.data_ptr<float>());
+        fda::wgrad_combine_launch(ws.data_ptr<float>(),
+                                  part.data_ptr<float>(),
+                                  (long)K * R * S * C, nch, cur_stream());
+        return ws;
+    }
     fda::conv_wgrad_launch(dy.data_ptr(), x.data_ptr(), ws.data_ptr<float>(),
                            N, H, W, C, K, P, Q, (int)R, (int)S, (int)sy,
                            (int)sx, (int)py, (int)px, cur_stream());
@@ -479,6 +495,22 @@ void conv_igemm_wgrad_into(at::Tensor dy, at::Tensor x, at::Tensor ws,
     const int K = (int)dy.size(1), P = (int)dy.size(2), Q = (int)dy.size(3);
     TORCH_CHECK(C % 64 == 0 && K % 64 == 0);
     TORCH_CHECK(ws.numel() == (int64_t)K * R * S * C);
+    if (fda::conv_wgrad_two_phase()) {
+        int ft, mch, nch;
+        fda::conv_wgrad_plan((long)N * P * Q, C, K, (int)R, (int)S,
+                             &ft, &mch, &nch);
+        auto part = at::empty({(int64_t)nch, (int64_t)K * R * S * C},
+                              x.options().dtype(at::kFloat));
+        fda::conv_wgrad_launch(dy.data_ptr(), x.data_ptr(),
+                               ws.data_ptr<float>(), N, H, W, C, K, P, Q,
+                               (int)R, (int)S, (int)sy, (int)sx, (int)py,
+                               (int)px, cur_stream(),
+                               part.data_ptr<float>());
+        fda::wgrad_combine_launch(ws.data_ptr<float>(),
+                                  part.data_ptr<float>(),
+                                  (long)K * R * S * C, nch, cur_stream());
+        return;
+    }
     fda::conv_wgrad_launch(dy.data_ptr(), x.data_ptr(), ws.data_ptr<float>(),
                            N, H, W, C, K, P, Q, (int)R, (int)S, (int)sy,
                            (int)sx, (int)py, (int)px, cur_stream());

@@ -1291,11 +1291,19 @@ typedef __attribute__((ext_vector_type(4))) short short4_;
 constexpr int WG_BM = 64;      // m per K-step
 constexpr int WG_MCH = 2048;   // pixels per block (chunk)
 
-template <int FT, bool STEM = false, bool ASMRD = false, int MB = WG_BM>
+template <int FT, bool STEM = false, bool ASMRD = false, int MB = WG_BM,
+          bool TP = false>
 __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
     const unsigned short* __restrict__ dy,   // [M][K] (NHWC out grad)
     const unsigned short* __restrict__ x,    // [N,H,W,C]
-    float* __restrict__ ws,                  // [K][RS*C] fp32, pre-zeroed
+    float* __restrict__ ws,  /* TP=false: [K][RS*C] fp32, pre-zeroed,
+                                accumulated with atomicAdd across chunks.
+                                TP=true (two-phase): [nch][K][RS*C] fp32
+                                partial slab, uninitialized — every chunk
+                                block plain-stores its full tile (PMC
+                                showed 100% of the atomics go to DRAM as
+                                serialized RMWs, docs/wgrad_study.md);
+                                wgrad_combine_kernel folds the slab. */
     int N, int H, int W, int C, int K, int P, int Q,
     int R, int S, int sy, int sx, int py, int px, int nch, int mch) {
     constexpr int TCH = 32 * FT;             // tile channels per operand
@@ -1571,6 +1579,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
     // epilogue: out[i=k][j=c]; C/D map col=lane&15, row=(lane>>4)*4+jj
     const int fcol = lane & 15, frow0 = (lane >> 4) * 4;
     const long RSC = (long)R * S * C;
+    float* const wsb = TP ? ws + (long)chunk * K * RSC : ws;
     #pragma unroll
     for (int ki = 0; ki < FT; ++ki)
         #pragma unroll
@@ -1579,9 +1588,39 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(
             for (int jj = 0; jj < 4; ++jj) {
                 const int kk = k0 + wk * 16 * FT + ki * 16 + frow0 + jj;
                 const int cc = c0 + wc * 16 * FT + ci * 16 + fcol;
-                atomicAdd(&ws[kk * RSC + (long)rs * C + cc],
-                          acc[ki][ci][jj]);
+                if constexpr (TP)
+                    wsb[kk * RSC + (long)rs * C + cc] = acc[ki][ci][jj];
+                else
+                    atomicAdd(&wsb[kk * RSC + (long)rs * C + cc],
+                              acc[ki][ci][jj]);
             }
+}
+
+// two-phase combine: ws[i] += sum_chunk part[chunk][i] (i over K*RSC,
+// float4-vectorized — C%64==0 so n%4==0). Streaming reads replace the
+// nch-deep DRAM atomic fan-in per address.
+__global__ __launch_bounds__(256) void wgrad_combine_kernel(
+    float* __restrict__ ws, const float* __restrict__ part, long n,
+    int nch) {
+    const long i = ((long)blockIdx.x * 256 + threadIdx.x) * 4;
+    if (i >= n) return;
+    floatx4 s = *(const floatx4*)(part + i);
+    for (int j = 1; j < nch; ++j) {
+        const floatx4 p = *(const floatx4*)(part + (long)j * n + i);
+        s.x += p.x; s.y += p.y; s.z += p.z; s.w += p.w;
+    }
+    floatx4* w = (floatx4*)(ws + i);
+    const floatx4 w0 = *w;
+    s.x += w0.x; s.y += w0.y; s.z += w0.z; s.w += w0.w;
+    *w = s;
+}
+
+void wgrad_combine_launch(float* ws, const float* part, long n, int nch,
+                          hipStream_t stream) {
+    const long lanes = n / 4;
+    dim3 grid((unsigned)((lanes + 255) / 256));
+    hipLaunchKernelGGL(wgrad_combine_kernel, grid, dim3(256), 0, stream,
+                       ws, part, n, nch);
 }
 
 static bool wgrad_asm() {
@@ -1634,11 +1673,11 @@ static int pick_mch(long M, long tiles, int target) {
     return (int)mch;
 }
 
-void conv_wgrad_launch(const void* dy, const void* x, float* ws,
-                       int N, int H, int W, int C, int K, int P, int Q,
-                       int R, int S, int sy, int sx, int py, int px,
-                       hipStream_t stream) {
-    const long M = (long)N * P * Q;
+// tile/chunk plan — the single source of truth shared with the bindings
+// (which size the two-phase partial slab from nch; same pattern as
+// conv_igemm_plan after the fwd-stats workspace near-miss).
+void conv_wgrad_plan(long M, int C, int K, int R, int S,
+                     int* ft, int* mch, int* nch) {
     // A/B: FLUXDIST_WGRAD_FT2=1 forces the 64x64 tile everywhere (its
     // 48 KB 3-ring fits 3 blocks/CU vs FT4's 2 — the BK32 lesson)
     static const bool force_ft2 = [] {
@@ -1650,8 +1689,37 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
     const long tiles4 = (long)(K / 128) * (C / 128) * R * S;
     const long blocks4_max = tiles4 * ((M + WG_MCH - 1) / WG_MCH);
     if (!force_ft2 && K % 128 == 0 && C % 128 == 0 && blocks4_max >= 192) {
-        const int mch = pick_mch(M, tiles4, 768);
-        const int nch = (int)((M + mch - 1) / mch);
+        *ft = 4;
+        *mch = pick_mch(M, tiles4, 768);
+    } else {
+        *ft = 2;
+        *mch = pick_mch(M, (long)(K / 64) * (C / 64) * R * S, 768);
+    }
+    *nch = (int)((M + *mch - 1) / *mch);
+}
+
+bool conv_wgrad_two_phase() {
+    // FLUXDIST_WGRAD_2PH=1: chunk blocks plain-store private partials,
+    // one combine kernel folds them (no DRAM atomics, no fill needed on
+    // the slab — docs/wgrad_study.md option 1). Mutually exclusive with
+    // the MB32 A/B knob, whose launch path ignores the slab — combining
+    // an unwritten slab would corrupt ws.
+    static const bool v = [] {
+        const char* e = getenv("FLUXDIST_WGRAD_2PH");
+        const char* m = getenv("FLUXDIST_WGRAD_MB32");
+        return e && e[0] == '1' && !(m && m[0] == '1');
+    }();
+    return v;
+}
+
+void conv_wgrad_launch(const void* dy, const void* x, float* ws,
+                       int N, int H, int W, int C, int K, int P, int Q,
+                       int R, int S, int sy, int sx, int py, int px,
+                       hipStream_t stream, float* part) {
+    const long M = (long)N * P * Q;
+    int FTp, mch, nch;
+    conv_wgrad_plan(M, C, K, R, S, &FTp, &mch, &nch);
+    if (FTp == 4) {
         dim3 grid((unsigned)(K / 128), (unsigned)(C / 128),
                   (unsigned)(R * S * nch));
         static const bool mb32 = [] {
@@ -1670,7 +1738,14 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
             return;
         }
         const size_t shmem = 2 * 2 * (WG_BM * 128) * sizeof(unsigned short);
-        if (wgrad_asm())
+        if (part)
+            hipLaunchKernelGGL((conv_wgrad_kernel<4, false, false, WG_BM,
+                                                  true>),
+                               grid, dim3(256), shmem, stream,
+                               (const unsigned short*)dy,
+                               (const unsigned short*)x, part, N, H, W, C,
+                               K, P, Q, R, S, sy, sx, py, px, nch, mch);
+        else if (wgrad_asm())
             hipLaunchKernelGGL((conv_wgrad_kernel<4, false, true>), grid,
                                dim3(256), shmem, stream,
                                (const unsigned short*)dy,
@@ -1682,13 +1757,17 @@ void conv_wgrad_launch(const void* dy, const void* x, float* ws,
                                (const unsigned short*)x, ws, N, H, W, C, K,
                                P, Q, R, S, sy, sx, py, px, nch, mch);
     } else {
-        const long tiles2 = (long)(K / 64) * (C / 64) * R * S;
-        const int mch = pick_mch(M, tiles2, 768);
-        const int nch = (int)((M + mch - 1) / mch);
         dim3 grid((unsigned)(K / 64), (unsigned)(C / 64),
                   (unsigned)(R * S * nch));
         const size_t shmem = 3 * 2 * (WG_BM * 64) * sizeof(unsigned short);
-        if (wgrad_asm())
+        if (part)
+            hipLaunchKernelGGL((conv_wgrad_kernel<2, false, false, WG_BM,
+                                                  true>),
+                               grid, dim3(256), shmem, stream,
+                               (const unsigned short*)dy,
+                               (const unsigned short*)x, part, N, H, W, C,
+                               K, P, Q, R, S, sy, sx, py, px, nch, mch);
+        else if (wgrad_asm())
             hipLaunchKernelGGL((conv_wgrad_kernel<2, false, true>), grid,
                                dim3(256), shmem, stream,
                                (const unsigned short*)dy,

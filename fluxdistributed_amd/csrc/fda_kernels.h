@@ -117,11 +117,20 @@ void conv_stem_wgrad_launch(const void* dy, const void* x, float* ws,
                             int N, int Hp, int Wp, int K, int P, int Q,
                             int R, int sy, int sx, hipStream_t stream);
 
-// wgrad: ws[K][RS*C] fp32 (pre-zeroed) += dy^T @ im2col(x), atomic chunks
+// wgrad: ws[K][RS*C] fp32 (pre-zeroed) += dy^T @ im2col(x), atomic chunks.
+// Two-phase mode (part != nullptr): chunk blocks plain-store into the
+// [nch][K][RS*C] slab `part` (uninitialized OK) instead of DRAM atomics;
+// follow with wgrad_combine_launch(ws, part, K*RS*C, nch).
 void conv_wgrad_launch(const void* dy, const void* x, float* ws,
                        int N, int H, int W, int C, int K, int P, int Q,
                        int R, int S, int sy, int sx, int py, int px,
-                       hipStream_t stream);
+                       hipStream_t stream, float* part = nullptr);
+// tile/chunk plan (single source of truth with the bindings): M = N*P*Q
+void conv_wgrad_plan(long M, int C, int K, int R, int S,
+                     int* ft, int* mch, int* nch);
+bool conv_wgrad_two_phase();   // FLUXDIST_WGRAD_2PH knob
+void wgrad_combine_launch(float* ws, const float* part, long n, int nch,
+                          hipStream_t stream);
 
 // batched conv-weight transpose: w[k][rc] -> wt[rc][k] for all tensors in
 // one launch (device pointer arrays)
