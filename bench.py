@@ -40,10 +40,13 @@ def parse_args():
     p.add_argument("--bucket-mb", type=float, default=25.0)
     p.add_argument("--no-overlap", action="store_true")
     p.add_argument("--graph", action="store_true",
-                   help="capture the step in a hipGraph (measured slower than "
-                        "eager on ResNet-34 bs96 — see profiles/README.md)")
+                   help="capture the step in a hipGraph (DEFAULT at world=1: "
+                        "+4.4%% same-box over eager once the loader's "
+                        "record_stream bug was fixed — docs/notes-round3.md)")
     p.add_argument("--no-graph", action="store_true",
-                   help="(default) eager step; kept for compatibility")
+                   help="force the eager step (the world>1 default: the "
+                        "bucketed all-reduce/backward overlap is hook-driven "
+                        "and stays on the validated eager path)")
     p.add_argument("--allow-cpu", action="store_true")
     p.add_argument("--lr", type=float, default=0.01)
     p.add_argument("--momentum", type=float, default=0.9)
@@ -119,7 +122,10 @@ def main():
 
     from fluxdistributed_amd.engine import make_train_step
 
-    use_graph = args.graph and not args.no_graph
+    # hipGraph is the single-GPU default (replay removes the ~23% of wall
+    # that is inter-kernel launch gaps, +4.4% same-box); world>1 keeps the
+    # eager step so the bucketed all-reduce overlaps backward via hooks.
+    use_graph = (args.graph or world == 1) and not args.no_graph
     example = next(loader) if (device.type == "cuda" and use_graph) else None
     train_step = make_train_step(model, opt, logit_cross_entropy,
                                  example_batch=example, ddp=ddp,
@@ -183,6 +189,8 @@ def main():
                 "image_size": args.image_size,
                 "num_classes": args.num_classes,
                 "parallelism": f"dp{max(world, 1)}",
+                "engine": ("graph" if (use_graph and device.type == "cuda")
+                           else "eager"),
                 "optimizer": f"sgd_momentum(lr={args.lr},m={args.momentum})",
                 "loss": "logitcrossentropy",
             },

@@ -67,7 +67,16 @@ class PrefetchLoader:
             yd = y.to(self.device, non_blocking=True)
         evt = torch.cuda.Event()
         evt.record(self.copy_stream)
-        torch.cuda.current_stream(self.device).wait_event(evt)
+        cur = torch.cuda.current_stream(self.device)
+        cur.wait_event(evt)
+        # xd/yd are ALLOCATED on copy_stream but consumed on the current
+        # stream: record_stream is the allocator contract for that — without
+        # it, freeing xd returns the block to copy_stream's pool and a later
+        # prefetch H2D can overwrite it while the consumer's kernels still
+        # read it. Latent under eager timing; exposed as input aliasing by
+        # back-to-back hipGraph replays (tools/graph_parity.py history).
+        xd.record_stream(cur)
+        yd.record_stream(cur)
         # pinned host tensors must outlive the async copy
         self._prev_batch = (x, y)
         return xd, yd
